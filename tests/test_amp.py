@@ -1,0 +1,80 @@
+"""Apex-style AMP: dynamic loss scaler dynamics, O2 master weights,
+scale_loss context, overflow skip — CPU (reference semantics)."""
+
+import torch
+
+from amdtrain.parallel import amp
+from amdtrain.parallel.amp import DynamicLossScaler
+
+
+def test_scaler_dynamics():
+    s = DynamicLossScaler(init_scale=1024.0, growth_interval=3, enabled=True)
+    s.update(found_inf=True)
+    assert s.scale == 512.0
+    s.update(False)
+    s.update(False)
+    assert s.scale == 512.0
+    s.update(False)  # 3rd clean step -> growth
+    assert s.scale == 1024.0
+    for _ in range(50):
+        s.update(True)
+    assert s.scale >= 1.0  # floor
+
+
+def _model_opt(dtype=torch.float32):
+    torch.manual_seed(0)
+    m = torch.nn.Sequential(torch.nn.Linear(4, 8), torch.nn.ReLU(),
+                            torch.nn.Linear(8, 2))
+    opt = torch.optim.SGD(m.parameters(), lr=0.1)
+    return m, opt
+
+
+def test_o1_initialize_keeps_fp32_params():
+    m, opt = _model_opt()
+    m2, opt2 = amp.initialize(m, opt, opt_level="O1", dtype=torch.bfloat16)
+    assert all(p.dtype == torch.float32 for p in m2.parameters())
+    x = torch.randn(3, 4)
+    loss = m2(x).sum()
+    with amp.scale_loss(loss, opt2) as scaled:
+        scaled.backward()
+    opt2.step()
+
+
+def test_o2_master_weights_flow():
+    m, opt = _model_opt()
+    m2, opt2 = amp.initialize(m, opt, opt_level="O2", dtype=torch.bfloat16)
+    # model params are half; optimizer steps fp32 masters
+    assert all(p.dtype == torch.bfloat16 for p in m2.parameters())
+    h = opt2._amp_handle
+    assert len(h.master_params) == len(list(m2.parameters()))
+    assert all(mp.dtype == torch.float32 for mp in h.master_params)
+    before = [p.detach().clone() for p in m2.parameters()]
+    x = torch.randn(3, 4).bfloat16()
+    loss = m2(x).float().sum()
+    with amp.scale_loss(loss, opt2) as scaled:
+        scaled.backward()
+    opt2.step()
+    after = [p.detach().clone() for p in m2.parameters()]
+    assert any(not torch.equal(a, b) for a, b in zip(before, after))
+    # masters and model params agree after the post-step cast
+    for mp, hp in zip(h.master_params, h.model_params):
+        assert torch.allclose(mp.bfloat16().float(), hp.float(), atol=1e-2)
+
+
+def test_overflow_skips_step():
+    m, opt = _model_opt()
+    m2, opt2 = amp.initialize(m, opt, opt_level="O1", dtype=torch.float16)
+    h = opt2._amp_handle
+    assert h.scaler.enabled  # fp16 needs scaling
+    before_scale = h.scaler.scale
+    x = torch.randn(3, 4)
+    loss = m2(x).sum() * float("inf")
+    with amp.scale_loss(loss, opt2) as scaled:
+        (scaled * 0 + m2(x).sum() * torch.tensor(float("nan"))).backward()
+    assert h.found_inf
+    assert h.scaler.scale == before_scale * 0.5
+    before = [p.detach().clone() for p in m2.parameters()]
+    opt2.step()  # must be skipped
+    assert h.steps_skipped == 1
+    for a, b in zip(before, m2.parameters()):
+        assert torch.equal(a, b)
