@@ -1,0 +1,466 @@
+// Fused NHWC BatchNorm(+residual add)(+ReLU) forward/backward for gfx950.
+//
+// The reference runs conv -> BN -> ReLU as separate cuDNN/elementwise
+// launches (53 BN layers in ResNet-50; SURVEY §2c "fuse into conv epilogue").
+// On MI355X the BN+add+ReLU tail is pure HBM traffic, so it runs as:
+//   fwd train: [reduce: per-channel sum/sumsq, fp32]  ->
+//              [finalize: mean/invstd + running-stat update + scale/shift] ->
+//              [apply: y = scale*x + shift (+z), relu — one pass]
+//   bwd:       [reduce: per-channel sum(ghat), sum(ghat*xhat)] ->
+//              [finalize: grad_w/grad_b + per-channel A,B,D coefficients] ->
+//              [apply: gx = A*ghat + B*x + D (+ optional ghat out)]
+//
+// Reduction strategy (CDNA4): rows R = N*H*W of C contiguous channels;
+// 16 B/lane vector loads; each thread owns a FIXED channel-group so partial
+// sums live in registers (no LDS atomics in the hot loop), then one LDS
+// phase-reduction per block and one global atomicAdd per channel per block.
+// All ResNet channel counts (64..2048) hit this register path.
+#include "common.h"
+
+namespace {
+
+// ---- reduction: per-channel (a, b) sums over rows -------------------------
+// fwd (BWD=false): a = x, b = x^2
+// bwd (BWD=true):  a = ghat, b = ghat * xhat,
+//                  ghat = go * (y > 0) when RELU else go
+template <typename T, int VEC, bool BWD, bool RELU>
+__global__ void __launch_bounds__(AMD_TPB)
+bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
+                 const T* __restrict__ y, const float* __restrict__ mean,
+                 const float* __restrict__ invstd, float* __restrict__ out,
+                 long R, int C) {
+  const int t = threadIdx.x;
+  const int gpr = C / VEC;  // channel-groups per row
+  // rows are split across blocks
+  const long rows_per_block = (R + gridDim.x - 1) / gridDim.x;
+  const long r0 = blockIdx.x * rows_per_block;
+  const long r1 = min(r0 + rows_per_block, R);
+
+  if (gpr <= AMD_TPB) {
+    // thread's channel group is fixed: gc = t % gpr; phase = t / gpr
+    const int gc = t % gpr;
+    const int phase = t / gpr;
+    const int rstep = AMD_TPB / gpr;
+    const int c0 = gc * VEC;
+    float sa[VEC], sb[VEC];
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) sa[k] = sb[k] = 0.f;
+    float mk[VEC], ik[VEC];
+    if (BWD) {
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        mk[k] = mean[c0 + k];
+        ik[k] = invstd[c0 + k];
+      }
+    }
+    for (long r = r0 + phase; r < r1; r += rstep) {
+      const long base = r * C + c0;
+      Pack<T, VEC> xv = *(const Pack<T, VEC>*)(x + base);
+      Pack<T, VEC> gv, yv;
+      if (BWD) {
+        gv = *(const Pack<T, VEC>*)(go + base);
+        if (RELU) yv = *(const Pack<T, VEC>*)(y + base);
+      }
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        float xe = to_f32(xv.v[k]);
+        if (BWD) {
+          float ge = to_f32(gv.v[k]);
+          if (RELU && to_f32(yv.v[k]) <= 0.f) ge = 0.f;
+          sa[k] += ge;
+          sb[k] += ge * (xe - mk[k]) * ik[k];
+        } else {
+          sa[k] += xe;
+          sb[k] += xe * xe;
+        }
+      }
+    }
+    // LDS reduce across phases, then one global atomicAdd per channel
+    __shared__ float lds[AMD_TPB * VEC];
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) lds[t * VEC + k] = sa[k];
+    __syncthreads();
+    if (phase == 0) {
+      for (int ph = 1; ph < rstep; ++ph)
+#pragma unroll
+        for (int k = 0; k < VEC; ++k)
+          sa[k] += lds[(ph * gpr + gc) * VEC + k];
+    }
+    __syncthreads();
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) lds[t * VEC + k] = sb[k];
+    __syncthreads();
+    if (phase == 0) {
+      for (int ph = 1; ph < rstep; ++ph)
+#pragma unroll
+        for (int k = 0; k < VEC; ++k)
+          sb[k] += lds[(ph * gpr + gc) * VEC + k];
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        atomicAdd(&out[c0 + k], sa[k]);
+        atomicAdd(&out[C + c0 + k], sb[k]);
+      }
+    }
+  } else {
+    // wide-C: each thread owns up to 2 channel-groups, no cross-thread reduce
+    const int cpt = gpr / AMD_TPB;  // host guarantees <= 2 and divisible
+    float sa[2][VEC], sb[2][VEC];
+    float mk[2][VEC], ik[2][VEC];
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        sa[j][k] = sb[j][k] = 0.f;
+        if (BWD && j < cpt) {
+          mk[j][k] = mean[(t + j * AMD_TPB) * VEC + k];
+          ik[j][k] = invstd[(t + j * AMD_TPB) * VEC + k];
+        }
+      }
+    for (long r = r0; r < r1; ++r) {
+      for (int j = 0; j < cpt; ++j) {
+        const int c0 = (t + j * AMD_TPB) * VEC;
+        const long base = r * C + c0;
+        Pack<T, VEC> xv = *(const Pack<T, VEC>*)(x + base);
+        Pack<T, VEC> gv, yv;
+        if (BWD) {
+          gv = *(const Pack<T, VEC>*)(go + base);
+          if (RELU) yv = *(const Pack<T, VEC>*)(y + base);
+        }
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) {
+          float xe = to_f32(xv.v[k]);
+          if (BWD) {
+            float ge = to_f32(gv.v[k]);
+            if (RELU && to_f32(yv.v[k]) <= 0.f) ge = 0.f;
+            sa[j][k] += ge;
+            sb[j][k] += ge * (xe - mk[j][k]) * ik[j][k];
+          } else {
+            sa[j][k] += xe;
+            sb[j][k] += xe * xe;
+          }
+        }
+      }
+    }
+    for (int j = 0; j < cpt; ++j) {
+      const int c0 = (t + j * AMD_TPB) * VEC;
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        atomicAdd(&out[c0 + k], sa[j][k]);
+        atomicAdd(&out[C + c0 + k], sb[j][k]);
+      }
+    }
+  }
+}
+
+// ---- finalize (train): stats + running update + scale/shift ---------------
+__global__ void bn_finalize_train_kernel(
+    const float* __restrict__ sums, const float* __restrict__ w,
+    const float* __restrict__ b, float* __restrict__ rm,
+    float* __restrict__ rv, float* __restrict__ mean,
+    float* __restrict__ invstd, float* __restrict__ scale,
+    float* __restrict__ shift, long R, int C, float momentum, float eps) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float m = sums[c] / R;
+  float var = fmaxf(sums[C + c] / R - m * m, 0.f);
+  float is = rsqrtf(var + eps);
+  mean[c] = m;
+  invstd[c] = is;
+  // torch semantics: running_var uses the unbiased estimator
+  float unbiased = R > 1 ? var * (float)R / (float)(R - 1) : var;
+  rm[c] = (1.f - momentum) * rm[c] + momentum * m;
+  rv[c] = (1.f - momentum) * rv[c] + momentum * unbiased;
+  float sc = w[c] * is;
+  scale[c] = sc;
+  shift[c] = b[c] - m * sc;
+}
+
+__global__ void bn_finalize_eval_kernel(
+    const float* __restrict__ w, const float* __restrict__ b,
+    const float* __restrict__ rm, const float* __restrict__ rv,
+    float* __restrict__ scale, float* __restrict__ shift, int C, float eps) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float is = rsqrtf(rv[c] + eps);
+  float sc = w[c] * is;
+  scale[c] = sc;
+  shift[c] = b[c] - rm[c] * sc;
+}
+
+// ---- apply: y = scale*x + shift (+z), relu --------------------------------
+template <typename T, int VEC, bool RELU, bool HAS_ADD>
+__global__ void __launch_bounds__(AMD_TPB)
+bn_apply_kernel(const T* __restrict__ x, const T* __restrict__ z,
+                T* __restrict__ y, const float* __restrict__ scale,
+                const float* __restrict__ shift, long total_vec, int C) {
+  extern __shared__ float sm[];  // [2][C]
+  const int gpr = C / VEC;
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    sm[c] = scale[c];
+    sm[C + c] = shift[c];
+  }
+  __syncthreads();
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total_vec;
+       i += (long)gridDim.x * blockDim.x) {
+    const int c0 = (int)(i % gpr) * VEC;
+    const long base = i * VEC;
+    Pack<T, VEC> xv = *(const Pack<T, VEC>*)(x + base);
+    Pack<T, VEC> zv;
+    if (HAS_ADD) zv = *(const Pack<T, VEC>*)(z + base);
+    Pack<T, VEC> yv;
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) {
+      float v = sm[c0 + k] * to_f32(xv.v[k]) + sm[C + c0 + k];
+      if (HAS_ADD) v += to_f32(zv.v[k]);
+      if (RELU) v = fmaxf(v, 0.f);
+      yv.v[k] = from_f32<T>(v);
+    }
+    *(Pack<T, VEC>*)(y + base) = yv;
+  }
+}
+
+// ---- bwd finalize: grad_w/grad_b + A,B,D coefficients ---------------------
+__global__ void bn_bwd_finalize_kernel(
+    const float* __restrict__ sums,  // [2][C]: sg, sgx
+    const float* __restrict__ w, const float* __restrict__ mean,
+    const float* __restrict__ invstd, float* __restrict__ gw,
+    float* __restrict__ gb, float* __restrict__ A, float* __restrict__ Bc,
+    float* __restrict__ Dc, long R, int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float sg = sums[c], sgx = sums[C + c];
+  gb[c] = sg;
+  gw[c] = sgx;
+  float a = w[c] * invstd[c];
+  float bcoef = -a * invstd[c] * sgx / R;
+  A[c] = a;
+  Bc[c] = bcoef;
+  Dc[c] = -a * sg / R - bcoef * mean[c];
+}
+
+// ---- bwd apply: gx = A*ghat + B*x + D -------------------------------------
+template <typename T, int VEC, bool RELU, bool WRITE_GHAT>
+__global__ void __launch_bounds__(AMD_TPB)
+bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict__ go,
+                    const T* __restrict__ y, T* __restrict__ gx,
+                    T* __restrict__ ghat_out, const float* __restrict__ A,
+                    const float* __restrict__ Bc, const float* __restrict__ Dc,
+                    long total_vec, int C) {
+  extern __shared__ float sm[];  // [3][C]
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    sm[c] = A[c];
+    sm[C + c] = Bc[c];
+    sm[2 * C + c] = Dc[c];
+  }
+  __syncthreads();
+  const int gpr = C / VEC;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total_vec;
+       i += (long)gridDim.x * blockDim.x) {
+    const int c0 = (int)(i % gpr) * VEC;
+    const long base = i * VEC;
+    Pack<T, VEC> xv = *(const Pack<T, VEC>*)(x + base);
+    Pack<T, VEC> gv = *(const Pack<T, VEC>*)(go + base);
+    Pack<T, VEC> yv;
+    if (RELU) yv = *(const Pack<T, VEC>*)(y + base);
+    Pack<T, VEC> out, gh;
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) {
+      float ge = to_f32(gv.v[k]);
+      if (RELU && to_f32(yv.v[k]) <= 0.f) ge = 0.f;
+      float v = sm[c0 + k] * ge + sm[C + c0 + k] * to_f32(xv.v[k]) +
+                sm[2 * C + c0 + k];
+      out.v[k] = from_f32<T>(v);
+      if (WRITE_GHAT) gh.v[k] = from_f32<T>(ge);
+    }
+    *(Pack<T, VEC>*)(gx + base) = out;
+    if (WRITE_GHAT) *(Pack<T, VEC>*)(ghat_out + base) = gh;
+  }
+}
+
+// ---- host helpers ---------------------------------------------------------
+
+static void check_nhwc(const at::Tensor& x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, "expected 4D CUDA tensor");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "expected channels_last (NHWC) tensor");
+}
+
+template <typename F>
+static void dispatch_vec(const at::Tensor& x, F fn) {
+  int C = (int)x.size(1);
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(), "bn",
+      [&] {
+        using devT = typename DevT<scalar_t>::type;
+        constexpr int VEC = 16 / sizeof(devT);
+        TORCH_CHECK(C % VEC == 0, "C must be a multiple of ", VEC);
+        TORCH_CHECK(C / VEC <= 2 * AMD_TPB, "C too large: ", C);
+        fn((devT*)nullptr, std::integral_constant<int, VEC>{});
+      });
+}
+
+static int bn_reduce_grid(long R, int C) {
+  // enough blocks to fill 256 CUs, but bounded so the atomic flush stays cheap
+  long rows_per_block = std::max<long>(1, R / 1024);
+  return (int)std::min<long>((R + rows_per_block - 1) / rows_per_block, 1024);
+}
+
+}  // namespace
+
+std::vector<at::Tensor> batch_norm_fwd_train(
+    at::Tensor x, at::Tensor weight, at::Tensor bias, at::Tensor running_mean,
+    at::Tensor running_var, double momentum, double eps, bool relu,
+    std::optional<at::Tensor> addend) {
+  check_nhwc(x);
+  const long N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const long R = N * H * W;
+  auto opts = x.options().dtype(at::kFloat);
+  auto sums = at::zeros({2, C}, opts);
+  auto mean = at::empty({C}, opts);
+  auto invstd = at::empty({C}, opts);
+  auto scale = at::empty({C}, opts);
+  auto shift = at::empty({C}, opts);
+  auto y = at::empty_like(x);
+  auto stream = at::cuda::getCurrentCUDAStream();
+
+  dispatch_vec(x, [&](auto* tp, auto vec) {
+    using devT = std::remove_pointer_t<decltype(tp)>;
+    constexpr int VEC = decltype(vec)::value;
+    int grid = bn_reduce_grid(R, C);
+    bn_reduce_kernel<devT, VEC, false, false>
+        <<<grid, AMD_TPB, 0, stream>>>((const devT*)x.const_data_ptr(),
+                                       nullptr, nullptr, nullptr, nullptr,
+                                       sums.data_ptr<float>(), R, (int)C);
+    CHECK_CUDA_OK();
+    int fgrid = (int)((C + AMD_TPB - 1) / AMD_TPB);
+    bn_finalize_train_kernel<<<fgrid, AMD_TPB, 0, stream>>>(
+        sums.data_ptr<float>(), weight.data_ptr<float>(),
+        bias.data_ptr<float>(), running_mean.data_ptr<float>(),
+        running_var.data_ptr<float>(), mean.data_ptr<float>(),
+        invstd.data_ptr<float>(), scale.data_ptr<float>(),
+        shift.data_ptr<float>(), R, (int)C, (float)momentum, (float)eps);
+    CHECK_CUDA_OK();
+    long total_vec = R * C / VEC;
+    int agrid = amd_grid(total_vec);
+    size_t smem = 2 * C * sizeof(float);
+    const devT* zp =
+        addend ? (const devT*)addend->const_data_ptr() : nullptr;
+#define APPLY(RELU_, ADD_)                                                  \
+  bn_apply_kernel<devT, VEC, RELU_, ADD_>                                   \
+      <<<agrid, AMD_TPB, smem, stream>>>(                                   \
+          (const devT*)x.const_data_ptr(), zp, (devT*)y.data_ptr(),         \
+          scale.data_ptr<float>(), shift.data_ptr<float>(),     \
+          total_vec, (int)C)
+    if (relu && addend) APPLY(true, true);
+    else if (relu) APPLY(true, false);
+    else if (addend) APPLY(false, true);
+    else APPLY(false, false);
+#undef APPLY
+    CHECK_CUDA_OK();
+  });
+  return {y, mean, invstd};
+}
+
+at::Tensor batch_norm_fwd_eval(at::Tensor x, at::Tensor weight,
+                               at::Tensor bias, at::Tensor running_mean,
+                               at::Tensor running_var, double eps, bool relu,
+                               std::optional<at::Tensor> addend) {
+  check_nhwc(x);
+  const long C = x.size(1);
+  const long R = x.numel() / C;
+  auto opts = x.options().dtype(at::kFloat);
+  auto scale = at::empty({C}, opts);
+  auto shift = at::empty({C}, opts);
+  auto y = at::empty_like(x);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  int fgrid = (int)((C + AMD_TPB - 1) / AMD_TPB);
+  bn_finalize_eval_kernel<<<fgrid, AMD_TPB, 0, stream>>>(
+      weight.data_ptr<float>(), bias.data_ptr<float>(),
+      running_mean.data_ptr<float>(), running_var.data_ptr<float>(),
+      scale.data_ptr<float>(), shift.data_ptr<float>(), (int)C, (float)eps);
+  CHECK_CUDA_OK();
+  dispatch_vec(x, [&](auto* tp, auto vec) {
+    using devT = std::remove_pointer_t<decltype(tp)>;
+    constexpr int VEC = decltype(vec)::value;
+    long total_vec = R * C / VEC;
+    int agrid = amd_grid(total_vec);
+    size_t smem = 2 * C * sizeof(float);
+    const devT* zp =
+        addend ? (const devT*)addend->const_data_ptr() : nullptr;
+#define APPLY(RELU_, ADD_)                                                  \
+  bn_apply_kernel<devT, VEC, RELU_, ADD_>                                   \
+      <<<agrid, AMD_TPB, smem, stream>>>(                                   \
+          (const devT*)x.const_data_ptr(), zp, (devT*)y.data_ptr(),         \
+          scale.data_ptr<float>(), shift.data_ptr<float>(),     \
+          total_vec, (int)C)
+    if (relu && addend) APPLY(true, true);
+    else if (relu) APPLY(true, false);
+    else if (addend) APPLY(false, true);
+    else APPLY(false, false);
+#undef APPLY
+    CHECK_CUDA_OK();
+  });
+  return y;
+}
+
+std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
+                                       at::Tensor y, at::Tensor weight,
+                                       at::Tensor mean, at::Tensor invstd,
+                                       bool relu) {
+  check_nhwc(x);
+  check_nhwc(grad_out);
+  const long C = x.size(1);
+  const long R = x.numel() / C;
+  auto opts = x.options().dtype(at::kFloat);
+  auto sums = at::zeros({2, C}, opts);
+  auto gw = at::empty({C}, opts);
+  auto gb = at::empty({C}, opts);
+  auto A = at::empty({C}, opts);
+  auto Bc = at::empty({C}, opts);
+  auto Dc = at::empty({C}, opts);
+  auto gx = at::empty_like(x);
+  at::Tensor ghat;
+  if (relu) ghat = at::empty_like(grad_out);
+  auto stream = at::cuda::getCurrentCUDAStream();
+
+  dispatch_vec(x, [&](auto* tp, auto vec) {
+    using devT = std::remove_pointer_t<decltype(tp)>;
+    constexpr int VEC = decltype(vec)::value;
+    int grid = bn_reduce_grid(R, C);
+#define REDUCE(RELU_)                                                       \
+  bn_reduce_kernel<devT, VEC, true, RELU_><<<grid, AMD_TPB, 0, stream>>>(   \
+      (const devT*)x.const_data_ptr(),                                      \
+      (const devT*)grad_out.const_data_ptr(),                               \
+      (const devT*)y.const_data_ptr(), mean.data_ptr<float>(),        \
+      invstd.data_ptr<float>(), sums.data_ptr<float>(), R, (int)C)
+    if (relu) REDUCE(true);
+    else REDUCE(false);
+#undef REDUCE
+    CHECK_CUDA_OK();
+    int fgrid = (int)((C + AMD_TPB - 1) / AMD_TPB);
+    bn_bwd_finalize_kernel<<<fgrid, AMD_TPB, 0, stream>>>(
+        sums.data_ptr<float>(), weight.data_ptr<float>(),
+        mean.data_ptr<float>(), invstd.data_ptr<float>(),
+        gw.data_ptr<float>(), gb.data_ptr<float>(), A.data_ptr<float>(),
+        Bc.data_ptr<float>(), Dc.data_ptr<float>(), R, (int)C);
+    CHECK_CUDA_OK();
+    long total_vec = R * C / VEC;
+    int agrid = amd_grid(total_vec);
+    size_t smem = 3 * C * sizeof(float);
+#define BAPPLY(RELU_, WG_)                                                  \
+  bn_bwd_apply_kernel<devT, VEC, RELU_, WG_>                                \
+      <<<agrid, AMD_TPB, smem, stream>>>(                                   \
+          (const devT*)x.const_data_ptr(),                                  \
+          (const devT*)grad_out.const_data_ptr(),                           \
+          (const devT*)y.const_data_ptr(), (devT*)gx.data_ptr(),            \
+          RELU_ ? (devT*)ghat.data_ptr() : nullptr,                         \
+          A.data_ptr<float>(), Bc.data_ptr<float>(),            \
+          Dc.data_ptr<float>(), total_vec, (int)C)
+    if (relu) BAPPLY(true, true);
+    else BAPPLY(false, false);
+#undef BAPPLY
+    CHECK_CUDA_OK();
+  });
+  if (!relu) ghat = grad_out;  // ghat == grad_out when no ReLU was fused
+  return {gx, gw, gb, ghat};
+}

@@ -1,0 +1,215 @@
+// Elementwise / reduction utility kernels for gfx950:
+//  * normalize_u8      — fused uint8->float cast + (x-mean)/std (the apex
+//                        prefetcher's GPU-side normalize,
+//                        apex_distributed.py:119-122,157-158)
+//  * topk_ranks        — rank of the label among [C] logits (accuracy(),
+//                        distributed.py:381-395, without materializing topk)
+//  * multi_tensor_scale_check — in-place scale + inf/nan detect (apex amp_C
+//                        unscale, apex_distributed.py:328-329)
+//  * multi_tensor_cast — fused dtype conversion (fp16/bf16 compression and
+//                        O2 master-weight copies, horovod_distributed.py:159)
+#include "common.h"
+
+namespace {
+
+// ---- normalize_u8: NHWC uint8 [N,H,W,3] -> float/bf16, 4 pixels/thread ----
+
+template <typename T>
+__global__ void __launch_bounds__(AMD_TPB)
+normalize_u8_kernel(const unsigned char* __restrict__ x, T* __restrict__ y,
+                    float m0, float m1, float m2, float i0, float i1, float i2,
+                    long npix) {
+  const float mean[3] = {m0, m1, m2};
+  const float inv[3] = {i0, i1, i2};
+  // 4 pixels = 12 consecutive bytes per thread
+  long nquad = npix / 4;
+  for (long q = (long)blockIdx.x * blockDim.x + threadIdx.x; q < nquad;
+       q += (long)gridDim.x * blockDim.x) {
+    long byte0 = q * 12;
+    // three aligned 4-byte loads
+    const uint32_t* xu = (const uint32_t*)(x + byte0);
+    uint32_t w0 = xu[0], w1 = xu[1], w2 = xu[2];
+    unsigned char b[12];
+    *(uint32_t*)(b + 0) = w0;
+    *(uint32_t*)(b + 4) = w1;
+    *(uint32_t*)(b + 8) = w2;
+#pragma unroll
+    for (int k = 0; k < 12; ++k) {
+      int c = k % 3;
+      y[byte0 + k] = from_f32<T>(((float)b[k] - mean[c]) * inv[c]);
+    }
+  }
+  // tail pixels
+  long tail = nquad * 4;
+  for (long pix = tail + (long)blockIdx.x * blockDim.x + threadIdx.x;
+       pix < npix; pix += (long)gridDim.x * blockDim.x) {
+#pragma unroll
+    for (int c = 0; c < 3; ++c) {
+      long idx = pix * 3 + c;
+      y[idx] = from_f32<T>(((float)x[idx] - mean[c]) * inv[c]);
+    }
+  }
+}
+
+// ---- topk_ranks: one wave per row --------------------------------------
+
+template <typename T>
+__global__ void __launch_bounds__(AMD_TPB)
+topk_ranks_kernel(const T* __restrict__ logits, const long* __restrict__ target,
+                  int* __restrict__ ranks, long B, long C) {
+  const int wave = threadIdx.x / AMD_WAVE;
+  const int lane = threadIdx.x % AMD_WAVE;
+  const long row = (long)blockIdx.x * (AMD_TPB / AMD_WAVE) + wave;
+  if (row >= B) return;
+  const T* x = logits + row * C;
+  const long tidx = target[row];
+  const float tv = to_f32(x[tidx]);
+  float cnt = 0.f;
+  for (long c = lane; c < C; c += AMD_WAVE) {
+    float v = to_f32(x[c]);
+    // rank rule: higher value first; ties broken by lower index
+    cnt += (v > tv || (v == tv && c < tidx)) ? 1.f : 0.f;
+  }
+  cnt = wave_reduce_sum(cnt);
+  if (lane == 0) ranks[row] = (int)cnt;
+}
+
+// ---- multi-tensor scale + inf/nan check --------------------------------
+
+template <typename T>
+__global__ void __launch_bounds__(AMD_TPB)
+mt_scale_kernel(MTMeta meta, float scale, float* __restrict__ found_inf) {
+  const int t = meta.t_for_block[blockIdx.x];
+  const long base = (long)meta.chunk_for_block[blockIdx.x] * MT_CHUNK;
+  const long end = min(base + MT_CHUNK, meta.sizes[t]);
+  T* __restrict__ p = (T*)meta.b[t];
+  bool bad = false;
+  for (long i = base + threadIdx.x; i < end; i += blockDim.x) {
+    float v = to_f32(p[i]) * scale;
+    bad |= !isfinite(v);
+    p[i] = from_f32<T>(v);
+  }
+  if (__any(bad) && (threadIdx.x % AMD_WAVE) == 0) *found_inf = 1.f;
+}
+
+// ---- multi-tensor cast --------------------------------------------------
+
+template <typename S, typename D>
+__global__ void __launch_bounds__(AMD_TPB)
+mt_cast_kernel(MTMeta meta) {
+  const int t = meta.t_for_block[blockIdx.x];
+  const long base = (long)meta.chunk_for_block[blockIdx.x] * MT_CHUNK;
+  const long end = min(base + MT_CHUNK, meta.sizes[t]);
+  const S* __restrict__ src = (const S*)meta.a[t];
+  D* __restrict__ dst = (D*)meta.b[t];
+  for (long i = base + threadIdx.x; i < end; i += blockDim.x) {
+    dst[i] = from_f32<D>(to_f32(src[i]));
+  }
+}
+
+}  // namespace
+
+at::Tensor normalize_u8(at::Tensor x, std::vector<double> mean,
+                        std::vector<double> std_, bool bf16_out) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kByte);
+  TORCH_CHECK(x.dim() == 4 && x.size(1) == 3, "expects NCHW with C=3");
+  auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
+  long npix = x.size(0) * x.size(2) * x.size(3);
+  auto opts = x.options().dtype(bf16_out ? at::kBFloat16 : at::kFloat);
+  auto y = at::empty(xc.sizes(),
+                     opts.memory_format(at::MemoryFormat::ChannelsLast));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  int grid = amd_grid(npix / 4 + 1);
+  float m[3], iv[3];
+  for (int c = 0; c < 3; ++c) {
+    m[c] = (float)mean[c];
+    iv[c] = 1.f / (float)std_[c];
+  }
+  if (bf16_out) {
+    normalize_u8_kernel<__hip_bfloat16><<<grid, AMD_TPB, 0, stream>>>(
+        (const unsigned char*)xc.const_data_ptr(),
+        (__hip_bfloat16*)y.data_ptr(), m[0], m[1], m[2], iv[0], iv[1], iv[2],
+        npix);
+  } else {
+    normalize_u8_kernel<float><<<grid, AMD_TPB, 0, stream>>>(
+        (const unsigned char*)xc.const_data_ptr(), y.data_ptr<float>(), m[0],
+        m[1], m[2], iv[0], iv[1], iv[2], npix);
+  }
+  CHECK_CUDA_OK();
+  return y;
+}
+
+at::Tensor topk_ranks(at::Tensor logits, at::Tensor target) {
+  TORCH_CHECK(logits.is_cuda() && logits.dim() == 2);
+  auto lc = logits.contiguous();
+  long B = lc.size(0), C = lc.size(1);
+  auto tl = target.to(at::kLong).contiguous();
+  auto ranks = at::empty({B}, lc.options().dtype(at::kInt));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  int rows_per_block = AMD_TPB / AMD_WAVE;
+  int grid = (int)((B + rows_per_block - 1) / rows_per_block);
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, lc.scalar_type(),
+      "topk_ranks", [&] {
+        using devT = typename DevT<scalar_t>::type;
+        topk_ranks_kernel<devT><<<grid, AMD_TPB, 0, stream>>>(
+            (const devT*)lc.const_data_ptr(), tl.data_ptr<long>(),
+            ranks.data_ptr<int>(), B, C);
+        CHECK_CUDA_OK();
+      });
+  return ranks;
+}
+
+void multi_tensor_scale_check(std::vector<at::Tensor> tensors, double scale,
+                              at::Tensor found_inf) {
+  if (tensors.empty()) return;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  // group by dtype (BN params stay fp32 in the O2 path -> mixed lists)
+  std::map<at::ScalarType, std::vector<at::Tensor>> groups;
+  for (auto& t : tensors) groups[t.scalar_type()].push_back(t);
+  for (auto& [st, group] : groups) {
+    AT_DISPATCH_FLOATING_TYPES_AND2(
+        at::ScalarType::BFloat16, at::ScalarType::Half, st, "mt_scale", [&] {
+          using devT = typename DevT<scalar_t>::type;
+          mt_apply(group, &group, nullptr,
+                   [&](const MTMeta& meta, int blocks, int) {
+                     mt_scale_kernel<devT><<<blocks, AMD_TPB, 0, stream>>>(
+                         meta, (float)scale, found_inf.data_ptr<float>());
+                     CHECK_CUDA_OK();
+                   });
+        });
+  }
+}
+
+void multi_tensor_cast(std::vector<at::Tensor> src,
+                       std::vector<at::Tensor> dst) {
+  TORCH_CHECK(src.size() == dst.size());
+  if (src.empty()) return;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  std::map<std::pair<at::ScalarType, at::ScalarType>,
+           std::pair<std::vector<at::Tensor>, std::vector<at::Tensor>>>
+      groups;
+  for (size_t i = 0; i < src.size(); ++i) {
+    auto key = std::make_pair(src[i].scalar_type(), dst[i].scalar_type());
+    groups[key].first.push_back(src[i]);
+    groups[key].second.push_back(dst[i]);
+  }
+  for (auto& [key, pair] : groups) {
+    AT_DISPATCH_FLOATING_TYPES_AND2(
+        at::ScalarType::BFloat16, at::ScalarType::Half, key.first, "mt_cast_s",
+        [&] {
+          using srcT = typename DevT<scalar_t>::type;
+          AT_DISPATCH_FLOATING_TYPES_AND2(
+              at::ScalarType::BFloat16, at::ScalarType::Half, key.second,
+              "mt_cast_d", [&] {
+                using dstT = typename DevT<scalar_t>::type;
+                mt_apply(pair.first, &pair.second, nullptr,
+                         [&](const MTMeta& meta, int blocks, int) {
+                           mt_cast_kernel<srcT, dstT>
+                               <<<blocks, AMD_TPB, 0, stream>>>(meta);
+                           CHECK_CUDA_OK();
+                         });
+              });
+        });
+  }
+}

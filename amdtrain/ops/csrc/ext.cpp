@@ -1,0 +1,72 @@
+// Python bindings for the amdtrain gfx950 HIP kernels (amdtrain._C).
+#include <torch/extension.h>
+
+#include <optional>
+#include <vector>
+
+// sgd.hip
+void multi_tensor_sgd(std::vector<at::Tensor> params,
+                      std::vector<at::Tensor> grads,
+                      std::vector<at::Tensor> momenta, double lr,
+                      double momentum, double weight_decay, bool nesterov,
+                      bool first, bool zero_grad);
+
+// cross_entropy.hip
+std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor target);
+at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor target,
+                             at::Tensor lse, double gscale);
+
+// elementwise.hip
+at::Tensor normalize_u8(at::Tensor x, std::vector<double> mean,
+                        std::vector<double> std_, bool bf16_out);
+at::Tensor topk_ranks(at::Tensor logits, at::Tensor target);
+void multi_tensor_scale_check(std::vector<at::Tensor> tensors, double scale,
+                              at::Tensor found_inf);
+void multi_tensor_cast(std::vector<at::Tensor> src,
+                       std::vector<at::Tensor> dst);
+
+// batchnorm.hip
+std::vector<at::Tensor> batch_norm_fwd_train(
+    at::Tensor x, at::Tensor weight, at::Tensor bias, at::Tensor running_mean,
+    at::Tensor running_var, double momentum, double eps, bool relu,
+    std::optional<at::Tensor> addend);
+at::Tensor batch_norm_fwd_eval(at::Tensor x, at::Tensor weight,
+                               at::Tensor bias, at::Tensor running_mean,
+                               at::Tensor running_var, double eps, bool relu,
+                               std::optional<at::Tensor> addend);
+std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
+                                       at::Tensor y, at::Tensor weight,
+                                       at::Tensor mean, at::Tensor invstd,
+                                       bool relu);
+
+// pool.hip
+std::vector<at::Tensor> max_pool_3x3_s2_fwd(at::Tensor x);
+at::Tensor max_pool_3x3_s2_bwd(at::Tensor grad_y, at::Tensor idx, long H,
+                               long W);
+at::Tensor global_avg_pool_fwd(at::Tensor x);
+at::Tensor global_avg_pool_bwd(at::Tensor grad_y, long H, long W);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "amdtrain hand-written gfx950 (CDNA4) HIP kernels";
+  m.def("multi_tensor_sgd", &multi_tensor_sgd, "fused multi-tensor SGD");
+  m.def("cross_entropy_fwd", &cross_entropy_fwd);
+  m.def("cross_entropy_bwd", &cross_entropy_bwd);
+  m.def("normalize_u8", &normalize_u8);
+  m.def("topk_ranks", &topk_ranks);
+  m.def("multi_tensor_scale_check", &multi_tensor_scale_check);
+  m.def("multi_tensor_cast", &multi_tensor_cast);
+  m.def("batch_norm_fwd_train", &batch_norm_fwd_train,
+        py::arg("x"), py::arg("weight"), py::arg("bias"),
+        py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
+        py::arg("eps"), py::arg("relu"),
+        py::arg("addend") = std::nullopt);
+  m.def("batch_norm_fwd_eval", &batch_norm_fwd_eval,
+        py::arg("x"), py::arg("weight"), py::arg("bias"),
+        py::arg("running_mean"), py::arg("running_var"), py::arg("eps"),
+        py::arg("relu"), py::arg("addend") = std::nullopt);
+  m.def("batch_norm_bwd", &batch_norm_bwd);
+  m.def("max_pool_3x3_s2_fwd", &max_pool_3x3_s2_fwd);
+  m.def("max_pool_3x3_s2_bwd", &max_pool_3x3_s2_bwd);
+  m.def("global_avg_pool_fwd", &global_avg_pool_fwd);
+  m.def("global_avg_pool_bwd", &global_avg_pool_bwd);
+}
