@@ -29,13 +29,16 @@ import torch.nn as nn
 from ..ops import fused as OF
 
 
+from ..ops.conv import AmdConv2d
+
+
 def conv3x3(in_planes: int, out_planes: int, stride: int = 1) -> nn.Conv2d:
-    return nn.Conv2d(in_planes, out_planes, kernel_size=3, stride=stride,
+    return AmdConv2d(in_planes, out_planes, kernel_size=3, stride=stride,
                      padding=1, bias=False)
 
 
 def conv1x1(in_planes: int, out_planes: int, stride: int = 1) -> nn.Conv2d:
-    return nn.Conv2d(in_planes, out_planes, kernel_size=1, stride=stride,
+    return AmdConv2d(in_planes, out_planes, kernel_size=1, stride=stride,
                      bias=False)
 
 

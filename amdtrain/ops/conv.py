@@ -1,0 +1,101 @@
+"""Hand-written 1x1 convolution over the gfx950 MFMA GEMM kernels.
+
+A 1x1 conv on NHWC data is exactly a GEMM over rows R = N*H*W
+(SURVEY §2c): forward C[R,Cout] = X[R,Cin] x W[Cout,Cin]^T, dgrad uses the
+pre-transposed weight, wgrad is the TN GEMM with split-M accumulation.
+Strided (downsample) 1x1 convs gather/scatter the even rows around the same
+GEMMs.
+
+Dispatch: ``AmdConv2d`` routes eligible 1x1 convs (GPU + extension + bf16
+path + AMDTRAIN_CONV1X1=custom) to these kernels, everything else to MIOpen
+via the standard conv path.  The default is chosen from measurement — see
+profiles/ for the per-shape comparison.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from ._ext import ext_available, require_ext
+
+
+def _rows(x: torch.Tensor) -> torch.Tensor:
+    """channels_last NCHW -> [N*H*W, C] view (no copy)."""
+    n, c, h, w = x.shape
+    return x.permute(0, 2, 3, 1).reshape(n * h * w, c)
+
+
+class _Conv1x1(torch.autograd.Function):
+    @staticmethod
+    @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
+    def forward(ctx, x: torch.Tensor, weight: torch.Tensor, stride: int):
+        e = require_ext()
+        n, cin, h, w = x.shape
+        cout = weight.shape[0]
+        xc = x.contiguous(memory_format=torch.channels_last)
+        if stride > 1:
+            xc = xc[:, :, ::stride, ::stride] \
+                .contiguous(memory_format=torch.channels_last)
+        ho, wo = xc.shape[2], xc.shape[3]
+        x2d = _rows(xc)
+        w2d = weight.reshape(cout, cin)
+        y2d = e.gemm_bt(x2d, w2d, False)
+        ctx.save_for_backward(x2d, w2d)
+        ctx.meta = (n, cin, h, w, stride, ho, wo, cout)
+        y = y2d.view(n, ho, wo, cout).permute(0, 3, 1, 2)
+        return y
+
+    @staticmethod
+    @torch.amp.custom_bwd(device_type="cuda")
+    def backward(ctx, grad_y: torch.Tensor):
+        e = require_ext()
+        x2d, w2d = ctx.saved_tensors
+        n, cin, h, w, stride, ho, wo, cout = ctx.meta
+        gy = grad_y.contiguous(memory_format=torch.channels_last)
+        gy2d = _rows(gy).to(torch.bfloat16)
+        # dgrad: dX = dY x W  (BT form with pre-transposed weight)
+        wT = e.transpose_2d(w2d)                      # [Cin, Cout]
+        dx2d = e.gemm_bt(gy2d, wT, False)             # [R_sub, Cin] bf16
+        dxs = dx2d.view(n, ho, wo, cin).permute(0, 3, 1, 2)
+        if stride > 1:
+            dx = torch.zeros(n, cin, h, w, dtype=dxs.dtype,
+                             device=dxs.device) \
+                .contiguous(memory_format=torch.channels_last)
+            dx[:, :, ::stride, ::stride] = dxs
+        else:
+            dx = dxs
+        # wgrad: dW[Cout,Cin] = dY^T x X (fp32 split-M accumulation)
+        dw = e.gemm_tn(gy2d, x2d, 0).reshape(cout, cin, 1, 1)
+        return dx, dw, None
+
+
+def conv1x1_mfma(x: torch.Tensor, weight: torch.Tensor,
+                 stride: int = 1) -> torch.Tensor:
+    return _Conv1x1.apply(x, weight, stride)
+
+
+def _conv1x1_env_default() -> str:
+    return os.environ.get("AMDTRAIN_CONV1X1", "custom")
+
+
+class AmdConv2d(nn.Conv2d):
+    """nn.Conv2d whose eligible 1x1 instances run the MFMA GEMM path.
+
+    State dict / init are identical to nn.Conv2d.  Non-1x1 shapes (3x3
+    spatial convs, the 7x7 stem) currently run through MIOpen; their
+    implicit-GEMM HIP kernels are the next build stage.
+    """
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if (self.kernel_size == (1, 1) and self.bias is None
+                and x.is_cuda and ext_available()
+                and _conv1x1_env_default() == "custom"
+                and self.in_channels % 32 == 0
+                and self.out_channels % 16 == 0
+                and self.padding == (0, 0)):
+            return conv1x1_mfma(x, self.weight, self.stride[0])
+        return super().forward(x)

@@ -39,6 +39,11 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
                                        at::Tensor mean, at::Tensor invstd,
                                        bool relu);
 
+// gemm.hip
+at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out);
+at::Tensor gemm_tn(at::Tensor dY, at::Tensor X, long msplit);
+at::Tensor transpose_2d(at::Tensor x);
+
 // pool.hip
 std::vector<at::Tensor> max_pool_3x3_s2_fwd(at::Tensor x);
 at::Tensor max_pool_3x3_s2_bwd(at::Tensor grad_y, at::Tensor idx, long H,
@@ -65,6 +70,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("running_mean"), py::arg("running_var"), py::arg("eps"),
         py::arg("relu"), py::arg("addend") = std::nullopt);
   m.def("batch_norm_bwd", &batch_norm_bwd);
+  m.def("gemm_bt", &gemm_bt, py::arg("A"), py::arg("B"),
+        py::arg("f32_out") = false);
+  m.def("gemm_tn", &gemm_tn, py::arg("dY"), py::arg("X"),
+        py::arg("msplit") = 0);
+  m.def("transpose_2d", &transpose_2d);
   m.def("max_pool_3x3_s2_fwd", &max_pool_3x3_s2_fwd);
   m.def("max_pool_3x3_s2_bwd", &max_pool_3x3_s2_bwd);
   m.def("global_avg_pool_fwd", &global_avg_pool_fwd);

@@ -1,0 +1,334 @@
+// MFMA bf16 GEMM kernels for gfx950 — the compute core of the hand-written
+// 1x1-convolution path ("Conv2d 1x1 (cuDNN) ... pointwise conv
+// == GEMM", ~36 call sites per ResNet-50 forward).
+//
+// Structure follows the measured CDNA4 recipe (cdna_hip_programming.md §5,
+// "m97" ladder step 3): 128x128 output tile, BK=32, 4 waves (2x2) per block,
+// each wave computing a 64x64 sub-tile as 4x4 fragments of
+// v_mfma_f32_16x16x32_bf16; A/B tiles staged to LDS with
+// __builtin_amdgcn_global_load_lds (16 B per lane, the direct-to-LDS DMA);
+// XCD-aware bijective blockIdx swizzle for L2 locality.
+//
+// Kernels:
+//   gemm_bt:  C[M,N] = A[M,K] x B[N,K]^T  (bf16 in, bf16 or f32 out)
+//             -> conv1x1 forward (A = NHWC activation rows, B = weight)
+//             -> conv1x1 dgrad   (A = grad rows, B = pre-transposed weight)
+//   gemm_tn:  C[N,K] += A[M,N]^T x B[M,K]  (f32 accumulation via split-M
+//             atomics) -> conv1x1 wgrad
+//   transpose_2d: bf16 [N,K] -> [K,N] (per-step weight transpose for dgrad)
+#include "common.h"
+
+namespace {
+
+using bf16 = __hip_bfloat16;
+using bf16x8 = __attribute__((ext_vector_type(8))) short;  // MFMA A/B frag
+using f32x4 = __attribute__((ext_vector_type(4))) float;   // MFMA C/D frag
+
+constexpr int BM = 128, BN = 128, BK = 32;
+constexpr int GEMM_TPB = 256;  // 4 waves, 2x2 wave grid, 64x64 per wave
+
+__device__ __forceinline__ short bf16_bits(bf16 v) {
+  short s;
+  __builtin_memcpy(&s, &v, 2);
+  return s;
+}
+
+// bijective XCD-aware swizzle (guide §5 "m204"): contiguous grid chunks map
+// to one XCD so neighboring tiles share that XCD's L2
+__device__ __forceinline__ int xcd_swizzle(int bid, int nwg) {
+  constexpr int NXCD = 8;
+  if (nwg < NXCD) return bid;
+  int xcd = bid % NXCD, idx = bid / NXCD;
+  int q = nwg / NXCD, r = nwg % NXCD;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+}
+
+// stage a BMxBK bf16 tile (rows `row0..row0+127` of a [Rows x ld] matrix,
+// cols k0..k0+31) into linear LDS [128][32] via global_load_lds.
+// 128*32*2B = 8 KiB = 512 lanes x 16 B -> 2 rounds of 256 threads.
+// Out-of-range rows are clamped (values unused, must just be finite).
+__device__ __forceinline__ void stage_tile_128x32(
+    const bf16* __restrict__ g, long ld, long row0, long rows, long k0,
+    bf16* lds) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int rnd = 0; rnd < 2; ++rnd) {
+    int unit = rnd * GEMM_TPB + t;       // 16B unit index 0..511
+    long row = row0 + (unit >> 2);       // 4 units per 64B row
+    if (row >= rows) row = rows - 1;     // clamp (finite garbage)
+    int koff = (unit & 3) * 8;           // 8 bf16 per 16B
+    const bf16* src = g + row * ld + k0 + koff;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)(lds + unit * 8), 16,
+        0, 0);
+  }
+}
+
+// C/D fragment mapping for mfma_f32_16x16x32_bf16 (guide §3, m89-verified):
+//   col = lane & 15, row = (lane >> 4) * 4 + reg
+// A fragment: lane holds A[row = lane&15][k = (lane>>4)*8 + j]
+// B fragment: lane holds B[k = (lane>>4)*8 + j][col = lane&15]
+//   (from LDS B^T tile [n][k] this is the same contiguous 16B read as A)
+
+template <bool F32OUT>
+__global__ void __launch_bounds__(GEMM_TPB, 2)
+gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
+               void* __restrict__ C, long M, long N, long K, int nbm,
+               int nbn) {
+  __shared__ bf16 As[BM * BK];
+  __shared__ bf16 Bs[BN * BK];
+
+  const int nwg = nbm * nbn;
+  const int bid = xcd_swizzle(blockIdx.x, nwg);
+  const int bm = bid / nbn, bn = bid % nbn;
+  const long m0 = (long)bm * BM, n0 = (long)bn * BN;
+
+  const int t = threadIdx.x;
+  const int wave = t / AMD_WAVE;      // 0..3 -> 2x2 wave grid
+  const int lane = t % AMD_WAVE;
+  const int wm = (wave >> 1) * 64;    // wave row offset in tile
+  const int wn = (wave & 1) * 64;
+  const int fr = lane & 15;           // fragment row/col within 16
+  const int fq = lane >> 4;           // quad index 0..3
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const long ksteps = K / BK;
+  for (long kt = 0; kt < ksteps; ++kt) {
+    __syncthreads();  // previous compute done before overwriting LDS
+    stage_tile_128x32(A, K, m0, M, kt * BK, As);
+    stage_tile_128x32(B, K, n0, N, kt * BK, Bs);
+    __syncthreads();  // barrier drains the global_load_lds queue
+
+    bf16x8 a[4], b[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      a[i] = *(const bf16x8*)&As[(wm + i * 16 + fr) * BK + fq * 8];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      b[j] = *(const bf16x8*)&Bs[(wn + j * 16 + fr) * BK + fq * 8];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a[i], b[j], acc[i][j], 0, 0, 0);
+  }
+
+  // epilogue: C[m0+wm+i*16+fq*4+r][n0+wn+j*16+fr]
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = m0 + wm + i * 16 + fq * 4 + r;
+        long col = n0 + wn + j * 16 + fr;
+        if (row < M && col < N) {
+          if (F32OUT)
+            ((float*)C)[row * N + col] = acc[i][j][r];
+          else
+            ((bf16*)C)[row * N + col] = __float2bfloat16(acc[i][j][r]);
+        }
+      }
+    }
+  }
+}
+
+// ---- TN GEMM for wgrad: dW[N,K] += sum_m dY[m,n] * X[m,k] ----------------
+// Tiles: output 128n x 128k, M chunked by 32 per K-step and split across
+// blocks (split-M) with fp32 atomic accumulation into dW.
+// LDS chunks are [32m][128c]; fragments need the m-major (transposed) read,
+// which is a strided ds_read per element (v1; tr_b16 upgrade is planned).
+__global__ void __launch_bounds__(GEMM_TPB, 2)
+gemm_tn_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
+               float* __restrict__ dW, long M, long N, long K, int nbn,
+               int nbk, int msplit) {
+  __shared__ bf16 Ys[32 * 128];
+  __shared__ bf16 Xs[32 * 128];
+
+  const int tiles = nbn * nbk;
+  const int tile = blockIdx.x % tiles;
+  const int mpart = blockIdx.x / tiles;
+  const int bn = tile / nbk, bk = tile % nbk;
+  const long n0 = (long)bn * BM, k0 = (long)bk * BN;
+
+  const long mchunks = (M + 31) / 32;
+  const long chunks_per_part = (mchunks + msplit - 1) / msplit;
+  const long mc0 = (long)mpart * chunks_per_part;
+  const long mc1 = min(mc0 + chunks_per_part, mchunks);
+
+  const int t = threadIdx.x;
+  const int wave = t / AMD_WAVE;
+  const int lane = t % AMD_WAVE;
+  const int wn = (wave >> 1) * 64;
+  const int wk = (wave & 1) * 64;
+  const int fr = lane & 15;
+  const int fq = lane >> 4;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (long mc = mc0; mc < mc1; ++mc) {
+    const long m0 = mc * 32;
+    __syncthreads();
+    // stage [32][128] chunks: 32 rows x 256B = 8 KiB each
+    {
+      const int tt = threadIdx.x;
+#pragma unroll
+      for (int rnd = 0; rnd < 2; ++rnd) {
+        int unit = rnd * GEMM_TPB + tt;  // 0..511, 16 units per row
+        long m = m0 + (unit >> 4);
+        if (m >= M) m = M - 1;
+        int coff = (unit & 15) * 8;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)(
+                dY + m * N + n0 + coff),
+            (__attribute__((address_space(3))) unsigned int*)(Ys + unit * 8),
+            16, 0, 0);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)(
+                X + m * K + k0 + coff),
+            (__attribute__((address_space(3))) unsigned int*)(Xs + unit * 8),
+            16, 0, 0);
+      }
+    }
+    __syncthreads();
+
+    const long mvalid = min((long)32, M - m0);
+    // A fragment: dY^T[n][m]: lane holds dY[m=fq*8+u][n0+wn+i*16+fr]
+    // B fragment: X[m=fq*8+u][k0+wk+j*16+fr]
+    bf16x8 a[4], b[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        int m = fq * 8 + u;
+        a[i][u] = bf16_bits(
+            m < mvalid ? Ys[m * 128 + wn + i * 16 + fr] : bf16(0.f));
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        int m = fq * 8 + u;
+        b[j][u] = bf16_bits(
+            m < mvalid ? Xs[m * 128 + wk + j * 16 + fr] : bf16(0.f));
+      }
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a[i], b[j], acc[i][j], 0, 0, 0);
+  }
+
+  // atomically accumulate into dW (split-M partials)
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long n = n0 + wn + i * 16 + fq * 4 + r;
+        long k = k0 + wk + j * 16 + fr;
+        if (n < N && k < K) atomicAdd(&dW[n * K + k], acc[i][j][r]);
+      }
+    }
+  }
+}
+
+// ---- small bf16 2D transpose (per-step weight transpose for dgrad) -------
+__global__ void __launch_bounds__(AMD_TPB)
+transpose_2d_kernel(const bf16* __restrict__ in, bf16* __restrict__ out,
+                    int R, int Ccols) {
+  __shared__ bf16 tile[32][33];
+  const int bx = blockIdx.x % ((Ccols + 31) / 32);
+  const int by = blockIdx.x / ((Ccols + 31) / 32);
+  const int c0 = bx * 32, r0 = by * 32;
+  const int tx = threadIdx.x % 32, ty = threadIdx.x / 32;  // 32x8
+  for (int dy = 0; dy < 32; dy += 8) {
+    int r = r0 + ty + dy, c = c0 + tx;
+    if (r < R && c < Ccols) tile[ty + dy][tx] = in[(long)r * Ccols + c];
+  }
+  __syncthreads();
+  for (int dy = 0; dy < 32; dy += 8) {
+    int c = c0 + ty + dy, r = r0 + tx;  // transposed coords
+    if (r < R && c < Ccols) out[(long)c * R + r] = tile[tx][ty + dy];
+  }
+}
+
+}  // namespace
+
+at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out) {
+  TORCH_CHECK(A.is_cuda() && A.dim() == 2 && B.dim() == 2);
+  TORCH_CHECK(A.scalar_type() == at::kBFloat16 &&
+              B.scalar_type() == at::kBFloat16,
+              "gemm_bt expects bf16 inputs");
+  auto Ac = A.contiguous();
+  auto Bc = B.contiguous();
+  long M = Ac.size(0), K = Ac.size(1), N = Bc.size(0);
+  TORCH_CHECK(Bc.size(1) == K, "K mismatch");
+  TORCH_CHECK(K % BK == 0, "K must be a multiple of ", BK);
+  auto C = at::empty({M, N},
+                     Ac.options().dtype(f32_out ? at::kFloat : at::kBFloat16));
+  int nbm = (int)((M + BM - 1) / BM), nbn = (int)((N + BN - 1) / BN);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (f32_out)
+    gemm_bt_kernel<true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+        (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
+        C.data_ptr(), M, N, K, nbm, nbn);
+  else
+    gemm_bt_kernel<false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+        (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
+        C.data_ptr(), M, N, K, nbm, nbn);
+  CHECK_CUDA_OK();
+  return C;
+}
+
+at::Tensor gemm_tn(at::Tensor dY, at::Tensor X, long msplit) {
+  TORCH_CHECK(dY.is_cuda() && dY.dim() == 2 && X.dim() == 2);
+  TORCH_CHECK(dY.scalar_type() == at::kBFloat16 &&
+              X.scalar_type() == at::kBFloat16);
+  auto Yc = dY.contiguous();
+  auto Xc = X.contiguous();
+  long M = Yc.size(0), N = Yc.size(1), K = Xc.size(1);
+  TORCH_CHECK(Xc.size(0) == M, "M mismatch");
+  auto dW = at::zeros({N, K}, Yc.options().dtype(at::kFloat));
+  int nbn = (int)((N + 127) / 128), nbk = (int)((K + 127) / 128);
+  if (msplit <= 0) {
+    // pick split so total blocks ~ 4 * 256 CUs
+    long tiles = (long)nbn * nbk;
+    msplit = std::max<long>(1, std::min<long>((M + 31) / 32, 1024 / tiles));
+  }
+  auto stream = at::cuda::getCurrentCUDAStream();
+  gemm_tn_kernel<<<(int)(nbn * nbk * msplit), GEMM_TPB, 0, stream>>>(
+      (const bf16*)Yc.const_data_ptr(), (const bf16*)Xc.const_data_ptr(),
+      dW.data_ptr<float>(), M, N, K, nbn, nbk, (int)msplit);
+  CHECK_CUDA_OK();
+  return dW;
+}
+
+at::Tensor transpose_2d(at::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 &&
+              x.scalar_type() == at::kBFloat16);
+  auto xc = x.contiguous();
+  int R = (int)xc.size(0), C = (int)xc.size(1);
+  auto out = at::empty({C, R}, xc.options());
+  int gx = (C + 31) / 32, gy = (R + 31) / 32;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  transpose_2d_kernel<<<gx * gy, AMD_TPB, 0, stream>>>(
+      (const bf16*)xc.const_data_ptr(), (bf16*)out.data_ptr(), R, C);
+  CHECK_CUDA_OK();
+  return out;
+}
