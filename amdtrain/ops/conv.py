@@ -78,6 +78,49 @@ def conv1x1_mfma(x: torch.Tensor, weight: torch.Tensor,
     return _Conv1x1.apply(x, weight, stride)
 
 
+class _Conv3x3(torch.autograd.Function):
+    """3x3 pad-1 conv (stride 1/2) over the implicit-GEMM gfx950 kernels."""
+
+    @staticmethod
+    @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
+    def forward(ctx, x: torch.Tensor, weight: torch.Tensor, stride: int):
+        e = require_ext()
+        n, cin, h, w = x.shape
+        cout = weight.shape[0]
+        xc = x.contiguous(memory_format=torch.channels_last)
+        x2d = _rows(xc)
+        # channels_last weight [Cout,Cin,3,3] is [Cout][kh][kw][Cin] in memory
+        w2d = weight.contiguous(memory_format=torch.channels_last) \
+            .permute(0, 2, 3, 1).reshape(cout, 9 * cin)
+        y2d = e.conv3x3_fwd(x2d, n, h, w, stride, w2d)
+        ctx.save_for_backward(x2d, w2d)
+        ctx.meta = (n, cin, h, w, stride, cout)
+        ho = (h + 2 - 3) // stride + 1
+        wo = (w + 2 - 3) // stride + 1
+        return y2d.view(n, ho, wo, cout).permute(0, 3, 1, 2)
+
+    @staticmethod
+    @torch.amp.custom_bwd(device_type="cuda")
+    def backward(ctx, grad_y: torch.Tensor):
+        e = require_ext()
+        x2d, w2d = ctx.saved_tensors
+        n, cin, h, w, stride, cout = ctx.meta
+        gy = grad_y.contiguous(memory_format=torch.channels_last)
+        gy2d = _rows(gy).to(torch.bfloat16)
+        dx2d = e.conv3x3_dgrad(gy2d, n, h, w, stride, w2d)
+        dx = dx2d.view(n, h, w, cin).permute(0, 3, 1, 2)
+        dw2d = e.conv3x3_wgrad(gy2d, x2d, n, h, w, stride)  # [Cout, 9*Cin] f32
+        # back to [Cout,Cin,3,3] (channels_last layout of the weight)
+        dw = dw2d.view(cout, 3, 3, cin).permute(0, 3, 1, 2) \
+            .contiguous(memory_format=torch.channels_last)
+        return dx, dw, None
+
+
+def conv3x3_mfma(x: torch.Tensor, weight: torch.Tensor,
+                 stride: int = 1) -> torch.Tensor:
+    return _Conv3x3.apply(x, weight, stride)
+
+
 def _conv1x1_env_default() -> str:
     return os.environ.get("AMDTRAIN_CONV1X1", "custom")
 
@@ -91,11 +134,15 @@ class AmdConv2d(nn.Conv2d):
     """
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        if (self.kernel_size == (1, 1) and self.bias is None
-                and x.is_cuda and ext_available()
-                and _conv1x1_env_default() == "custom"
+        if (x.is_cuda and ext_available() and self.bias is None
                 and self.in_channels % 32 == 0
-                and self.out_channels % 16 == 0
-                and self.padding == (0, 0)):
-            return conv1x1_mfma(x, self.weight, self.stride[0])
+                and self.out_channels % 16 == 0):
+            if (self.kernel_size == (1, 1) and self.padding == (0, 0)
+                    and _conv1x1_env_default() == "custom"):
+                return conv1x1_mfma(x, self.weight, self.stride[0])
+            if (self.kernel_size == (3, 3) and self.padding == (1, 1)
+                    and self.stride[0] in (1, 2)
+                    and os.environ.get("AMDTRAIN_CONV3X3", "custom")
+                    == "custom"):
+                return conv3x3_mfma(x, self.weight, self.stride[0])
         return super().forward(x)

@@ -1,0 +1,372 @@
+// Implicit-GEMM 3x3 convolution (NHWC, pad 1, stride 1/2) for gfx950 —
+// forward, dgrad and wgrad (SURVEY §2c "Conv2d 3x3 ... 16 call sites/fwd").
+//
+// GEMM view over rows m = (n, ho, wo):
+//   fwd:   Y[m, cout] = sum_{tap, cin} X[gather(m, tap), cin] * W[cout, tap, cin]
+//   dgrad: dX[m, cin] = sum_{tap, cout} dY[gather'(m, tap), cout] * W'[cin, tap, cout]
+//          (W' is the 180-degree-rotated, [Cin, 9*Cout]-permuted weight)
+//   wgrad: dW[cout, tap, cin] = sum_m dY[m, cout] * X[gather(m, tap), cin]
+//          (one split-M TN GEMM per tap, atomic fp32 accumulation)
+//
+// Same block structure as gemm.hip (128x128 tile, BK=32 within one tap —
+// all ResNet channel counts are multiples of 32 — 4 waves of 4x4
+// mfma_f32_16x16x32_bf16 fragments, global_load_lds staging).  Padding is
+// handled with a 16 B zero page: out-of-bounds gather rows load from it, so
+// no LDS pre-zeroing pass and no boundary branches in the MFMA loop.
+#include "common.h"
+
+namespace {
+
+using bf16 = __hip_bfloat16;
+using bf16x8 = __attribute__((ext_vector_type(8))) short;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+constexpr int GEMM_TPB = 256;
+constexpr int BK = 32;
+
+__device__ __forceinline__ int xcd_swz(int bid, int nwg) {
+  constexpr int NXCD = 8;
+  if (nwg < NXCD) return bid;
+  int xcd = bid % NXCD, idx = bid / NXCD;
+  int q = nwg / NXCD, r = nwg % NXCD;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+}
+
+struct ConvGeom {
+  int H, W, Hout, Wout, stride;  // input / output spatial dims
+};
+
+// forward gather: output row m=(n,ho,wo), tap (kh,kw) -> input row or -1
+__device__ __forceinline__ long fwd_gather(long m, int kh, int kw,
+                                           const ConvGeom& g) {
+  long t = m;
+  const int wo = (int)(t % g.Wout); t /= g.Wout;
+  const int ho = (int)(t % g.Hout); t /= g.Hout;
+  const int h = ho * g.stride - 1 + kh;
+  const int w = wo * g.stride - 1 + kw;
+  if (h < 0 || h >= g.H || w < 0 || w >= g.W) return -1;
+  return (t * g.H + h) * g.W + w;
+}
+
+// dgrad gather: input row m=(n,h,w), tap (kh,kw) -> dY row or -1
+// (H,W = input dims; Hout,Wout = dY dims)
+__device__ __forceinline__ long dgrad_gather(long m, int kh, int kw,
+                                             const ConvGeom& g) {
+  long t = m;
+  const int w = (int)(t % g.W); t /= g.W;
+  const int h = (int)(t % g.H); t /= g.H;
+  const int ho2 = h + 1 - kh, wo2 = w + 1 - kw;
+  if (g.stride == 1) {
+    if (ho2 < 0 || ho2 >= g.Hout || wo2 < 0 || wo2 >= g.Wout) return -1;
+    return (t * g.Hout + ho2) * g.Wout + wo2;
+  }
+  if ((ho2 & 1) || (wo2 & 1)) return -1;
+  const int ho = ho2 >> 1, wo = wo2 >> 1;
+  if (ho < 0 || ho >= g.Hout || wo < 0 || wo >= g.Wout) return -1;
+  return (t * g.Hout + ho) * g.Wout + wo;
+}
+
+// stage a [128 rows][32 ch] tile of gathered rows into linear LDS
+template <bool DGRAD>
+__device__ __forceinline__ void stage_gathered(
+    const bf16* __restrict__ src, int ld, long m0, long M, int c0, int kh,
+    int kw, const ConvGeom& g, const bf16* __restrict__ zero_page,
+    bf16* lds) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int rnd = 0; rnd < 2; ++rnd) {
+    int unit = rnd * GEMM_TPB + t;   // 0..511; 4 x 16B units per row
+    long m = m0 + (unit >> 2);
+    if (m >= M) m = M - 1;
+    long row = DGRAD ? dgrad_gather(m, kh, kw, g) : fwd_gather(m, kh, kw, g);
+    const bf16* p = row < 0 ? zero_page
+                            : src + row * (long)ld + c0 + (unit & 3) * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)p,
+        (__attribute__((address_space(3))) unsigned int*)(lds + unit * 8), 16,
+        0, 0);
+  }
+}
+
+// stage a [128 rows][32 cols] tile of a plain [Rows x ld] matrix at column
+// offset koff (weights)
+__device__ __forceinline__ void stage_plain(
+    const bf16* __restrict__ gsrc, long ld, long row0, long rows, long koff,
+    bf16* lds) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int rnd = 0; rnd < 2; ++rnd) {
+    int unit = rnd * GEMM_TPB + t;
+    long row = row0 + (unit >> 2);
+    if (row >= rows) row = rows - 1;
+    const bf16* src = gsrc + row * ld + koff + (unit & 3) * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)(lds + unit * 8), 16,
+        0, 0);
+  }
+}
+
+// fwd / dgrad main kernel.  DGRAD only changes the gather map; operand roles:
+//   fwd:   A = x rows (AC channels), B = w [NC, 9*AC], C = y [M, NC]
+//   dgrad: A = dy rows (AC = Cout), B = w' [NC = Cin, 9*Cout], C = dx
+template <bool DGRAD>
+__global__ void __launch_bounds__(GEMM_TPB, 2)
+conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
+               bf16* __restrict__ C, long M, int AC, int NC, ConvGeom g,
+               int nbm, int nbn, const bf16* __restrict__ zero_page) {
+  __shared__ bf16 As[128 * BK];
+  __shared__ bf16 Bs[128 * BK];
+
+  const int bid = xcd_swz(blockIdx.x, nbm * nbn);
+  const int bm = bid / nbn, bn = bid % nbn;
+  const long m0 = (long)bm * 128, n0 = (long)bn * 128;
+
+  const int t = threadIdx.x;
+  const int wave = t / AMD_WAVE, lane = t % AMD_WAVE;
+  const int wm = (wave >> 1) * 64, wn = (wave & 1) * 64;
+  const int fr = lane & 15, fq = lane >> 4;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int ksteps_per_tap = AC / BK;
+  for (int tap = 0; tap < 9; ++tap) {
+    const int kh = tap / 3, kw = tap % 3;
+    for (int ks = 0; ks < ksteps_per_tap; ++ks) {
+      const int c0 = ks * BK;
+      __syncthreads();
+      stage_gathered<DGRAD>(A, AC, m0, M, c0, kh, kw, g, zero_page, As);
+      stage_plain(Bw, (long)9 * AC, n0, NC, (long)tap * AC + c0, Bs);
+      __syncthreads();
+
+      bf16x8 a[4], b[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        a[i] = *(const bf16x8*)&As[(wm + i * 16 + fr) * BK + fq * 8];
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        b[j] = *(const bf16x8*)&Bs[(wn + j * 16 + fr) * BK + fq * 8];
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+  }
+
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = m0 + wm + i * 16 + fq * 4 + r;
+        long col = n0 + wn + j * 16 + fr;
+        if (row < M && col < NC)
+          C[row * NC + col] = __float2bfloat16(acc[i][j][r]);
+      }
+}
+
+// wgrad per tap: dW[cout, tap*Cin + cin] += sum_m dY[m, cout] * Xg[m, cin]
+__global__ void __launch_bounds__(GEMM_TPB, 2)
+conv3x3_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
+                     float* __restrict__ dW, long M, int Cout, int Cin,
+                     ConvGeom g, int tap, int nbn, int nbk, int msplit,
+                     const bf16* __restrict__ zero_page) {
+  __shared__ bf16 Ys[32 * 128];
+  __shared__ bf16 Xs[32 * 128];
+
+  const int kh = tap / 3, kw = tap % 3;
+  const int tiles = nbn * nbk;
+  const int tile = blockIdx.x % tiles;
+  const int mpart = blockIdx.x / tiles;
+  const int bn = tile / nbk, bk = tile % nbk;
+  const long n0 = (long)bn * 128, k0 = (long)bk * 128;
+
+  const long mchunks = (M + 31) / 32;
+  const long cpp = (mchunks + msplit - 1) / msplit;
+  const long mc0 = (long)mpart * cpp;
+  const long mc1 = min(mc0 + cpp, mchunks);
+
+  const int t = threadIdx.x;
+  const int wave = t / AMD_WAVE, lane = t % AMD_WAVE;
+  const int wn = (wave >> 1) * 64, wk = (wave & 1) * 64;
+  const int fr = lane & 15, fq = lane >> 4;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (long mc = mc0; mc < mc1; ++mc) {
+    const long m0 = mc * 32;
+    __syncthreads();
+#pragma unroll
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      int unit = rnd * GEMM_TPB + t;  // 16 units per 256B row
+      long m = m0 + (unit >> 4);
+      if (m >= M) m = M - 1;
+      int coff = (unit & 15) * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              dY + m * Cout + n0 + coff),
+          (__attribute__((address_space(3))) unsigned int*)(Ys + unit * 8),
+          16, 0, 0);
+      long xrow = fwd_gather(m, kh, kw, g);
+      const bf16* xp = xrow < 0 ? zero_page : X + xrow * (long)Cin + k0 + coff;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)xp,
+          (__attribute__((address_space(3))) unsigned int*)(Xs + unit * 8),
+          16, 0, 0);
+    }
+    __syncthreads();
+
+    const long mvalid = min((long)32, M - m0);
+    bf16x8 a[4], b[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        int m = fq * 8 + u;
+        bf16 v = m < mvalid ? Ys[m * 128 + wn + i * 16 + fr] : bf16(0.f);
+        short s;
+        __builtin_memcpy(&s, &v, 2);
+        a[i][u] = s;
+      }
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        int m = fq * 8 + u;
+        bf16 v = m < mvalid ? Xs[m * 128 + wk + j * 16 + fr] : bf16(0.f);
+        short s;
+        __builtin_memcpy(&s, &v, 2);
+        b[j][u] = s;
+      }
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a[i], b[j], acc[i][j], 0, 0, 0);
+  }
+
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long n = n0 + wn + i * 16 + fq * 4 + r;   // cout
+        long k = k0 + wk + j * 16 + fr;           // cin
+        if (n < Cout && k < Cin)
+          atomicAdd(&dW[n * (long)(9 * Cin) + (long)tap * Cin + k],
+                    acc[i][j][r]);
+      }
+}
+
+// permute weight [Cout, 9, Cin] -> rotated [Cin, 9, Cout] for dgrad:
+// W'[cin][tap][cout] = W[cout][8 - tap][cin]
+__global__ void __launch_bounds__(AMD_TPB)
+rotate_weight_kernel(const bf16* __restrict__ w, bf16* __restrict__ out,
+                     int Cout, int Cin) {
+  long total = (long)Cout * 9 * Cin;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long t = i;
+    const int cin = (int)(t % Cin); t /= Cin;
+    const int tap = (int)(t % 9); t /= 9;
+    const int cout = (int)t;
+    out[((long)cin * 9 + tap) * Cout + cout] =
+        w[((long)cout * 9 + (8 - tap)) * Cin + cin];
+  }
+}
+
+}  // namespace
+
+// host-side 16B zero page (device memory), one per device, created lazily
+static at::Tensor zero_page_for(const at::Tensor& like) {
+  static thread_local at::Tensor zp;
+  if (!zp.defined() || zp.device() != like.device())
+    zp = at::zeros({16}, like.options().dtype(at::kBFloat16));
+  return zp;
+}
+
+at::Tensor conv3x3_fwd(at::Tensor x2d, long Nn, long H, long W, long stride,
+                       at::Tensor w2d) {
+  // x2d: [Nn*H*W, Cin] bf16 NHWC rows; w2d: [Cout, 9*Cin]
+  TORCH_CHECK(x2d.is_cuda() && x2d.scalar_type() == at::kBFloat16);
+  long Cin = x2d.size(1), Cout = w2d.size(0);
+  TORCH_CHECK(w2d.size(1) == 9 * Cin);
+  TORCH_CHECK(Cin % BK == 0, "Cin must be multiple of 32");
+  long Hout = (H + 2 - 3) / stride + 1, Wout = (W + 2 - 3) / stride + 1;
+  long M = Nn * Hout * Wout;
+  auto y = at::empty({M, Cout}, x2d.options());
+  ConvGeom g{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
+  int nbm = (int)((M + 127) / 128), nbn = (int)((Cout + 127) / 128);
+  auto zp = zero_page_for(x2d);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  conv3x3_kernel<false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+      (const bf16*)x2d.const_data_ptr(), (const bf16*)w2d.const_data_ptr(),
+      (bf16*)y.data_ptr(), M, (int)Cin, (int)Cout, g, nbm, nbn,
+      (const bf16*)zp.const_data_ptr());
+  CHECK_CUDA_OK();
+  return y;
+}
+
+at::Tensor conv3x3_dgrad(at::Tensor dy2d, long Nn, long H, long W,
+                         long stride, at::Tensor w2d) {
+  // dy2d: [Nn*Hout*Wout, Cout]; returns dx2d [Nn*H*W, Cin]
+  long Cout = dy2d.size(1), Cin = w2d.size(1) / 9;
+  TORCH_CHECK(w2d.size(0) == Cout && w2d.size(1) == 9 * Cin);
+  TORCH_CHECK(Cout % BK == 0);
+  long Hout = (H + 2 - 3) / stride + 1, Wout = (W + 2 - 3) / stride + 1;
+  long M = Nn * H * W;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  // build rotated-permuted weight W' [Cin, 9*Cout]
+  auto wrot = at::empty({Cin, 9 * Cout}, w2d.options());
+  rotate_weight_kernel<<<amd_grid(Cout * 9 * Cin), AMD_TPB, 0, stream>>>(
+      (const bf16*)w2d.const_data_ptr(), (bf16*)wrot.data_ptr(), (int)Cout,
+      (int)Cin);
+  CHECK_CUDA_OK();
+  auto dx = at::empty({M, Cin}, dy2d.options());
+  ConvGeom g{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
+  int nbm = (int)((M + 127) / 128), nbn = (int)((Cin + 127) / 128);
+  auto zp = zero_page_for(dy2d);
+  conv3x3_kernel<true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+      (const bf16*)dy2d.const_data_ptr(), (const bf16*)wrot.const_data_ptr(),
+      (bf16*)dx.data_ptr(), M, (int)Cout, (int)Cin, g, nbm, nbn,
+      (const bf16*)zp.const_data_ptr());
+  CHECK_CUDA_OK();
+  return dx;
+}
+
+at::Tensor conv3x3_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn, long H,
+                         long W, long stride) {
+  long Cout = dy2d.size(1), Cin = x2d.size(1);
+  long Hout = (H + 2 - 3) / stride + 1, Wout = (W + 2 - 3) / stride + 1;
+  long M = Nn * Hout * Wout;
+  TORCH_CHECK(dy2d.size(0) == M);
+  auto dW = at::zeros({Cout, 9 * Cin}, dy2d.options().dtype(at::kFloat));
+  ConvGeom g{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
+  int nbn = (int)((Cout + 127) / 128), nbk = (int)((Cin + 127) / 128);
+  long tiles = (long)nbn * nbk;
+  int msplit = (int)std::max<long>(
+      1, std::min<long>((M + 31) / 32, 1024 / tiles));
+  auto zp = zero_page_for(x2d);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  for (int tap = 0; tap < 9; ++tap) {
+    conv3x3_wgrad_kernel<<<(int)(tiles * msplit), GEMM_TPB, 0, stream>>>(
+        (const bf16*)dy2d.const_data_ptr(), (const bf16*)x2d.const_data_ptr(),
+        dW.data_ptr<float>(), M, (int)Cout, (int)Cin, g, tap, nbn, nbk,
+        msplit, (const bf16*)zp.const_data_ptr());
+    CHECK_CUDA_OK();
+  }
+  return dW;
+}

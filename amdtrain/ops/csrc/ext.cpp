@@ -44,6 +44,14 @@ at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out);
 at::Tensor gemm_tn(at::Tensor dY, at::Tensor X, long msplit);
 at::Tensor transpose_2d(at::Tensor x);
 
+// conv3x3.hip
+at::Tensor conv3x3_fwd(at::Tensor x2d, long Nn, long H, long W, long stride,
+                       at::Tensor w2d);
+at::Tensor conv3x3_dgrad(at::Tensor dy2d, long Nn, long H, long W,
+                         long stride, at::Tensor w2d);
+at::Tensor conv3x3_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn, long H,
+                         long W, long stride);
+
 // pool.hip
 std::vector<at::Tensor> max_pool_3x3_s2_fwd(at::Tensor x);
 at::Tensor max_pool_3x3_s2_bwd(at::Tensor grad_y, at::Tensor idx, long H,
@@ -75,6 +83,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_tn", &gemm_tn, py::arg("dY"), py::arg("X"),
         py::arg("msplit") = 0);
   m.def("transpose_2d", &transpose_2d);
+  m.def("conv3x3_fwd", &conv3x3_fwd);
+  m.def("conv3x3_dgrad", &conv3x3_dgrad);
+  m.def("conv3x3_wgrad", &conv3x3_wgrad);
   m.def("max_pool_3x3_s2_fwd", &max_pool_3x3_s2_fwd);
   m.def("max_pool_3x3_s2_bwd", &max_pool_3x3_s2_bwd);
   m.def("global_avg_pool_fwd", &global_avg_pool_fwd);
