@@ -66,19 +66,61 @@ __device__ __forceinline__ long dgrad_gather(long m, int kh, int kw,
   return (t * g.Hout + ho) * g.Wout + wo;
 }
 
-// stage a [128 rows][32 ch] tile of gathered rows into linear LDS
+// per-unit precomputed pixel coordinates: the output-row decode (div/mod
+// chains) is tap-invariant, so it is hoisted out of the K-loop entirely
+struct UnitCoord {
+  long n_off;  // n * H * W (input-row base for this image)
+  int hb, wb;  // tap-0 pixel coords (may be negative / out of range)
+};
+
+template <bool DGRAD>
+__device__ __forceinline__ UnitCoord decode_unit(long m, const ConvGeom& g) {
+  UnitCoord u;
+  long t = m;
+  if (DGRAD) {  // m ranges over INPUT pixels; gather reads dY [Hout,Wout]
+    const int w = (int)(t % g.W); t /= g.W;
+    const int h = (int)(t % g.H); t /= g.H;
+    u.n_off = t * (long)g.Hout * g.Wout;
+    u.hb = h + 1;  // minus kh per tap
+    u.wb = w + 1;
+  } else {      // m ranges over OUTPUT pixels; gather reads X [H,W]
+    const int wo = (int)(t % g.Wout); t /= g.Wout;
+    const int ho = (int)(t % g.Hout); t /= g.Hout;
+    u.n_off = t * (long)g.H * g.W;
+    u.hb = ho * g.stride - 1;  // plus kh per tap
+    u.wb = wo * g.stride - 1;
+  }
+  return u;
+}
+
+template <bool DGRAD>
+__device__ __forceinline__ long unit_row(const UnitCoord& u, int kh, int kw,
+                                         const ConvGeom& g) {
+  if (DGRAD) {
+    int ho2 = u.hb - kh, wo2 = u.wb - kw;
+    if (g.stride == 2) {
+      if ((ho2 | wo2) & 1) return -1;
+      ho2 >>= 1;
+      wo2 >>= 1;
+    }
+    if (ho2 < 0 || ho2 >= g.Hout || wo2 < 0 || wo2 >= g.Wout) return -1;
+    return u.n_off + (long)ho2 * g.Wout + wo2;
+  }
+  const int h = u.hb + kh, w = u.wb + kw;
+  if (h < 0 || h >= g.H || w < 0 || w >= g.W) return -1;
+  return u.n_off + (long)h * g.W + w;
+}
+
 template <bool DGRAD>
 __device__ __forceinline__ void stage_gathered(
-    const bf16* __restrict__ src, int ld, long m0, long M, int c0, int kh,
+    const bf16* __restrict__ src, int ld, const UnitCoord* uc, int c0, int kh,
     int kw, const ConvGeom& g, const bf16* __restrict__ zero_page,
     bf16* lds) {
   const int t = threadIdx.x;
 #pragma unroll
   for (int rnd = 0; rnd < 2; ++rnd) {
     int unit = rnd * GEMM_TPB + t;   // 0..511; 4 x 16B units per row
-    long m = m0 + (unit >> 2);
-    if (m >= M) m = M - 1;
-    long row = DGRAD ? dgrad_gather(m, kh, kw, g) : fwd_gather(m, kh, kw, g);
+    long row = unit_row<DGRAD>(uc[rnd], kh, kw, g);
     const bf16* p = row < 0 ? zero_page
                             : src + row * (long)ld + c0 + (unit & 3) * 8;
     __builtin_amdgcn_global_load_lds(
@@ -133,13 +175,22 @@ conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+  // hoist the per-unit coordinate decode out of the K-loop
+  UnitCoord uc[2];
+#pragma unroll
+  for (int rnd = 0; rnd < 2; ++rnd) {
+    long m = m0 + ((rnd * GEMM_TPB + t) >> 2);
+    if (m >= M) m = M - 1;
+    uc[rnd] = decode_unit<DGRAD>(m, g);
+  }
+
   const int ksteps_per_tap = AC / BK;
   for (int tap = 0; tap < 9; ++tap) {
     const int kh = tap / 3, kw = tap % 3;
     for (int ks = 0; ks < ksteps_per_tap; ++ks) {
       const int c0 = ks * BK;
       __syncthreads();
-      stage_gathered<DGRAD>(A, AC, m0, M, c0, kh, kw, g, zero_page, As);
+      stage_gathered<DGRAD>(A, AC, uc, c0, kh, kw, g, zero_page, As);
       stage_plain(Bw, (long)9 * AC, n0, NC, (long)tap * AC + c0, Bs);
       __syncthreads();
 
