@@ -15,7 +15,7 @@ from typing import Optional
 import torch
 import torch.nn as nn
 
-from ..comm.collectives import broadcast_module_state, _broadcast_coalesced
+from ..comm.collectives import broadcast_module_state
 from ..comm.init import is_distributed
 from .reducer import BucketedReducer
 
@@ -35,16 +35,40 @@ class NativeDDP(nn.Module):
         self.reducer = BucketedReducer(
             list(module.parameters()), bucket_cap_mb=bucket_cap_mb,
             compression=compression, process_group=process_group)
-        self._buffers_list = [b for b in module.buffers()
-                              if b.dtype.is_floating_point or
-                              b.dtype in (torch.int64, torch.int32)]
+        self._flat_buffers = self._flatten_buffers(module)
+
+    @staticmethod
+    @torch.no_grad()
+    def _flatten_buffers(module):
+        """Re-home same-dtype module buffers (BN running stats etc.) as views
+        into one flat tensor per dtype, so the per-forward rank-0 broadcast
+        is a single RCCL collective with no pack/unpack copies."""
+        groups = {}
+        for mod in module.modules():
+            for name, buf in list(mod._buffers.items()):
+                if buf is None or buf.numel() == 0:
+                    continue
+                groups.setdefault(buf.dtype, []).append((mod, name, buf))
+        flats = []
+        for dtype, items in groups.items():
+            flat = torch.empty(sum(b.numel() for _, _, b in items),
+                               dtype=dtype, device=items[0][2].device)
+            off = 0
+            for mod, name, buf in items:
+                n = buf.numel()
+                view = flat[off:off + n].view_as(buf)
+                view.copy_(buf)
+                mod._buffers[name] = view
+                off += n
+            flats.append(flat)
+        return flats
 
     def forward(self, *args, **kwargs):
-        if (self.training and self.broadcast_buffers and is_distributed()
-                and self._buffers_list):
+        if self.training and self.broadcast_buffers and is_distributed():
             # rank-0 buffer broadcast each forward (torch-DDP semantics for
-            # BN running stats), coalesced into one flat RCCL broadcast
-            _broadcast_coalesced(self._buffers_list, src=0)
+            # BN running stats) — one flat broadcast per dtype
+            for flat in self._flat_buffers:
+                torch.distributed.broadcast(flat, src=0)
         return self.module(*args, **kwargs)
 
     def zero_grad(self, set_to_none: bool = False):  # type: ignore[override]
