@@ -39,6 +39,8 @@ def parse_args():
     p.add_argument("--image-size", type=int, default=224)
     p.add_argument("--no-ext", action="store_true",
                    help="A/B: plain-PyTorch ops instead of the HIP kernels")
+    p.add_argument("--profile", type=str, default="",
+                   help="write a torch.profiler kernel table here (3 steps)")
     return p.parse_args()
 
 
@@ -103,6 +105,12 @@ def main():
 
     for _ in range(args.warmup):
         step()
+    if args.profile and rank == 0:
+        from amdtrain.utils.profiling import profile_steps
+        with profile_steps(args.profile):
+            for _ in range(3):
+                step()
+        torch.cuda.synchronize()
     comm.barrier()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
