@@ -134,7 +134,11 @@ class AmdConv2d(nn.Conv2d):
     """
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        if (x.is_cuda and ext_available() and self.bias is None
+        # MFMA path is bf16: eligible under autocast or with bf16 inputs;
+        # fp32 inference/training falls through to MIOpen
+        bf16_ok = (x.dtype == torch.bfloat16
+                   or torch.is_autocast_enabled("cuda"))
+        if (x.is_cuda and bf16_ok and ext_available() and self.bias is None
                 and self.in_channels % 32 == 0
                 and self.out_channels % 16 == 0):
             if (self.kernel_size == (1, 1) and self.padding == (0, 0)

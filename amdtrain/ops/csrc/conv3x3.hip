@@ -264,13 +264,16 @@ conv3x3_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
       long m = m0 + (unit >> 4);
       if (m >= M) m = M - 1;
       int coff = (unit & 15) * 8;
+      const bf16* yp = (n0 + coff + 8 <= Cout) ? dY + m * Cout + n0 + coff
+                                               : zero_page;
       __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int*)(
-              dY + m * Cout + n0 + coff),
+          (const __attribute__((address_space(1))) unsigned int*)yp,
           (__attribute__((address_space(3))) unsigned int*)(Ys + unit * 8),
           16, 0, 0);
       long xrow = fwd_gather(m, kh, kw, g);
-      const bf16* xp = xrow < 0 ? zero_page : X + xrow * (long)Cin + k0 + coff;
+      const bf16* xp = (xrow < 0 || k0 + coff + 8 > Cin)
+                           ? zero_page
+                           : X + xrow * (long)Cin + k0 + coff;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)xp,
           (__attribute__((address_space(3))) unsigned int*)(Xs + unit * 8),

@@ -148,7 +148,7 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 gemm_tn_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
                float* __restrict__ dW, long M, long N, long K, int nbn,
-               int nbk, int msplit) {
+               int nbk, int msplit, const bf16* __restrict__ zero_page) {
   __shared__ bf16 Ys[32 * 128];
   __shared__ bf16 Xs[32 * 128];
 
@@ -189,14 +189,18 @@ gemm_tn_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
         long m = m0 + (unit >> 4);
         if (m >= M) m = M - 1;
         int coff = (unit & 15) * 8;
+        // rows can be narrower than the 128-col chunk (N or K < 128):
+        // out-of-row columns read the zero page instead of faulting
+        const bf16* yp = (n0 + coff + 8 <= N) ? dY + m * N + n0 + coff
+                                              : zero_page;
+        const bf16* xp = (k0 + coff + 8 <= K) ? X + m * K + k0 + coff
+                                              : zero_page;
         __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int*)(
-                dY + m * N + n0 + coff),
+            (const __attribute__((address_space(1))) unsigned int*)yp,
             (__attribute__((address_space(3))) unsigned int*)(Ys + unit * 8),
             16, 0, 0);
         __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int*)(
-                X + m * K + k0 + coff),
+            (const __attribute__((address_space(1))) unsigned int*)xp,
             (__attribute__((address_space(3))) unsigned int*)(Xs + unit * 8),
             16, 0, 0);
       }
@@ -311,10 +315,12 @@ at::Tensor gemm_tn(at::Tensor dY, at::Tensor X, long msplit) {
     long tiles = (long)nbn * nbk;
     msplit = std::max<long>(1, std::min<long>((M + 31) / 32, 1024 / tiles));
   }
+  auto zp = at::zeros({16}, Yc.options());
   auto stream = at::cuda::getCurrentCUDAStream();
   gemm_tn_kernel<<<(int)(nbn * nbk * msplit), GEMM_TPB, 0, stream>>>(
       (const bf16*)Yc.const_data_ptr(), (const bf16*)Xc.const_data_ptr(),
-      dW.data_ptr<float>(), M, N, K, nbn, nbk, (int)msplit);
+      dW.data_ptr<float>(), M, N, K, nbn, nbk, (int)msplit,
+      (const bf16*)zp.const_data_ptr());
   CHECK_CUDA_OK();
   return dW;
 }
