@@ -94,6 +94,10 @@ def run_worker(local_rank: int, nprocs: int, args, style: StyleConfig,
     if args.resume:
         ck = load_checkpoint(args.resume, model)
         args.start_epoch = ck.get("epoch", args.start_epoch)
+    elif args.pretrained and os.path.isfile("checkpoint.pth.tar"):
+        # no model-zoo download in this environment: --pretrained loads the
+        # local checkpoint (reference loads torchvision zoo weights here)
+        load_checkpoint("checkpoint.pth.tar", model)
 
     train_loader, val_loader, train_sampler, _ = build_loaders(
         args, world_size=world, rank=rank,
