@@ -95,10 +95,13 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
 #pragma unroll
         for (int k = 0; k < VEC; ++k)
           sb[k] += lds[(ph * gpr + gc) * VEC + k];
+      // per-block partials (no global atomics: 1024-way same-line atomic
+      // contention was ~24 ms/step — two-stage instead)
+      float* part = out + (long)blockIdx.x * 2 * C;
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
-        atomicAdd(&out[c0 + k], sa[k]);
-        atomicAdd(&out[C + c0 + k], sb[k]);
+        part[c0 + k] = sa[k];
+        part[C + c0 + k] = sb[k];
       }
     }
   } else {
@@ -141,12 +144,13 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
         }
       }
     }
+    float* part = out + (long)blockIdx.x * 2 * C;
     for (int j = 0; j < cpt; ++j) {
       const int c0 = (t + j * AMD_TPB) * VEC;
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
-        atomicAdd(&out[c0 + k], sa[j][k]);
-        atomicAdd(&out[C + c0 + k], sb[j][k]);
+        part[c0 + k] = sa[j][k];
+        part[C + c0 + k] = sb[j][k];
       }
     }
   }
@@ -154,15 +158,21 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
 
 // ---- finalize (train): stats + running update + scale/shift ---------------
 __global__ void bn_finalize_train_kernel(
-    const float* __restrict__ sums, const float* __restrict__ w,
+    const float* __restrict__ partials, int nparts,
+    const float* __restrict__ w,
     const float* __restrict__ b, float* __restrict__ rm,
     float* __restrict__ rv, float* __restrict__ mean,
     float* __restrict__ invstd, float* __restrict__ scale,
     float* __restrict__ shift, long R, int C, float momentum, float eps) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float m = sums[c] / R;
-  float var = fmaxf(sums[C + c] / R - m * m, 0.f);
+  float s1 = 0.f, s2 = 0.f;
+  for (int p = 0; p < nparts; ++p) {
+    s1 += partials[(long)p * 2 * C + c];
+    s2 += partials[(long)p * 2 * C + C + c];
+  }
+  float m = s1 / R;
+  float var = fmaxf(s2 / R - m * m, 0.f);
   float is = rsqrtf(var + eps);
   mean[c] = m;
   invstd[c] = is;
@@ -221,14 +231,18 @@ bn_apply_kernel(const T* __restrict__ x, const T* __restrict__ z,
 
 // ---- bwd finalize: grad_w/grad_b + A,B,D coefficients ---------------------
 __global__ void bn_bwd_finalize_kernel(
-    const float* __restrict__ sums,  // [2][C]: sg, sgx
+    const float* __restrict__ partials, int nparts,
     const float* __restrict__ w, const float* __restrict__ mean,
     const float* __restrict__ invstd, float* __restrict__ gw,
     float* __restrict__ gb, float* __restrict__ A, float* __restrict__ Bc,
     float* __restrict__ Dc, long R, int C) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float sg = sums[c], sgx = sums[C + c];
+  float sg = 0.f, sgx = 0.f;
+  for (int p = 0; p < nparts; ++p) {
+    sg += partials[(long)p * 2 * C + c];
+    sgx += partials[(long)p * 2 * C + C + c];
+  }
   gb[c] = sg;
   gw[c] = sgx;
   float a = w[c] * invstd[c];
@@ -300,9 +314,9 @@ static void dispatch_vec(const at::Tensor& x, F fn) {
 }
 
 static int bn_reduce_grid(long R, int C) {
-  // enough blocks to fill 256 CUs, but bounded so the atomic flush stays cheap
-  long rows_per_block = std::max<long>(1, R / 1024);
-  return (int)std::min<long>((R + rows_per_block - 1) / rows_per_block, 1024);
+  // enough blocks to saturate HBM reads; partial-buffer cost is grid*2C f32
+  long rows_per_block = std::max<long>(1, R / 512);
+  return (int)std::min<long>((R + rows_per_block - 1) / rows_per_block, 512);
 }
 
 }  // namespace
@@ -315,7 +329,8 @@ std::vector<at::Tensor> batch_norm_fwd_train(
   const long N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   const long R = N * H * W;
   auto opts = x.options().dtype(at::kFloat);
-  auto sums = at::zeros({2, C}, opts);
+  const int rgrid = bn_reduce_grid(R, C);
+  auto sums = at::empty({rgrid, 2 * C}, opts);
   auto mean = at::empty({C}, opts);
   auto invstd = at::empty({C}, opts);
   auto scale = at::empty({C}, opts);
@@ -326,15 +341,14 @@ std::vector<at::Tensor> batch_norm_fwd_train(
   dispatch_vec(x, [&](auto* tp, auto vec) {
     using devT = std::remove_pointer_t<decltype(tp)>;
     constexpr int VEC = decltype(vec)::value;
-    int grid = bn_reduce_grid(R, C);
     bn_reduce_kernel<devT, VEC, false, false>
-        <<<grid, AMD_TPB, 0, stream>>>((const devT*)x.const_data_ptr(),
-                                       nullptr, nullptr, nullptr, nullptr,
-                                       sums.data_ptr<float>(), R, (int)C);
+        <<<rgrid, AMD_TPB, 0, stream>>>((const devT*)x.const_data_ptr(),
+                                        nullptr, nullptr, nullptr, nullptr,
+                                        sums.data_ptr<float>(), R, (int)C);
     CHECK_CUDA_OK();
     int fgrid = (int)((C + AMD_TPB - 1) / AMD_TPB);
     bn_finalize_train_kernel<<<fgrid, AMD_TPB, 0, stream>>>(
-        sums.data_ptr<float>(), weight.data_ptr<float>(),
+        sums.data_ptr<float>(), rgrid, weight.data_ptr<float>(),
         bias.data_ptr<float>(), running_mean.data_ptr<float>(),
         running_var.data_ptr<float>(), mean.data_ptr<float>(),
         invstd.data_ptr<float>(), scale.data_ptr<float>(),
@@ -412,7 +426,8 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
   const long C = x.size(1);
   const long R = x.numel() / C;
   auto opts = x.options().dtype(at::kFloat);
-  auto sums = at::zeros({2, C}, opts);
+  const int rgrid = bn_reduce_grid(R, C);
+  auto sums = at::empty({rgrid, 2 * C}, opts);
   auto gw = at::empty({C}, opts);
   auto gb = at::empty({C}, opts);
   auto A = at::empty({C}, opts);
@@ -426,9 +441,8 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
   dispatch_vec(x, [&](auto* tp, auto vec) {
     using devT = std::remove_pointer_t<decltype(tp)>;
     constexpr int VEC = decltype(vec)::value;
-    int grid = bn_reduce_grid(R, C);
 #define REDUCE(RELU_)                                                       \
-  bn_reduce_kernel<devT, VEC, true, RELU_><<<grid, AMD_TPB, 0, stream>>>(   \
+  bn_reduce_kernel<devT, VEC, true, RELU_><<<rgrid, AMD_TPB, 0, stream>>>(  \
       (const devT*)x.const_data_ptr(),                                      \
       (const devT*)grad_out.const_data_ptr(),                               \
       (const devT*)y.const_data_ptr(), mean.data_ptr<float>(),        \
@@ -439,7 +453,7 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
     CHECK_CUDA_OK();
     int fgrid = (int)((C + AMD_TPB - 1) / AMD_TPB);
     bn_bwd_finalize_kernel<<<fgrid, AMD_TPB, 0, stream>>>(
-        sums.data_ptr<float>(), weight.data_ptr<float>(),
+        sums.data_ptr<float>(), rgrid, weight.data_ptr<float>(),
         mean.data_ptr<float>(), invstd.data_ptr<float>(),
         gw.data_ptr<float>(), gb.data_ptr<float>(), A.data_ptr<float>(),
         Bc.data_ptr<float>(), Dc.data_ptr<float>(), R, (int)C);

@@ -224,13 +224,18 @@ conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
 }
 
 // wgrad per tap: dW[cout, tap*Cin + cin] += sum_m dY[m, cout] * Xg[m, cin]
+// Transposed reg-staging + contiguous b128 fragment reads (see gemm.hip
+// gemm_tn); gathered X rows that fall in the padding stage zeros.
+__device__ __forceinline__ int wg_swz(int cc, int m) {
+  return cc * 32 + (m ^ (((cc >> 3) & 3) << 3));
+}
+
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 conv3x3_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
                      float* __restrict__ dW, long M, int Cout, int Cin,
-                     ConvGeom g, int tap, int nbn, int nbk, int msplit,
-                     const bf16* __restrict__ zero_page) {
-  __shared__ bf16 Ys[32 * 128];
-  __shared__ bf16 Xs[32 * 128];
+                     ConvGeom g, int tap, int nbn, int nbk, int msplit) {
+  __shared__ bf16 Ys[128 * 32];
+  __shared__ bf16 Xs[128 * 32];
 
   const int kh = tap / 3, kw = tap % 3;
   const int tiles = nbn * nbk;
@@ -260,49 +265,43 @@ conv3x3_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
     __syncthreads();
 #pragma unroll
     for (int rnd = 0; rnd < 2; ++rnd) {
-      int unit = rnd * GEMM_TPB + t;  // 16 units per 256B row
-      long m = m0 + (unit >> 4);
-      if (m >= M) m = M - 1;
-      int coff = (unit & 15) * 8;
-      const bf16* yp = (n0 + coff + 8 <= Cout) ? dY + m * Cout + n0 + coff
-                                               : zero_page;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int*)yp,
-          (__attribute__((address_space(3))) unsigned int*)(Ys + unit * 8),
-          16, 0, 0);
-      long xrow = fwd_gather(m, kh, kw, g);
-      const bf16* xp = (xrow < 0 || k0 + coff + 8 > Cin)
-                           ? zero_page
-                           : X + xrow * (long)Cin + k0 + coff;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int*)xp,
-          (__attribute__((address_space(3))) unsigned int*)(Xs + unit * 8),
-          16, 0, 0);
+      const int unit = rnd * GEMM_TPB + t;  // 16 units (8 cols) per m-row
+      const long m = m0 + (unit >> 4);
+      const int c0 = (unit & 15) * 8;
+      const int mloc = unit >> 4;
+      // dY chunk (cols may exceed Cout for Cout<128 -> zeros)
+      bf16 yv[8];
+      if (m < M && n0 + c0 + 8 <= Cout) {
+        uint4 raw = *(const uint4*)(dY + m * Cout + n0 + c0);
+        __builtin_memcpy(yv, &raw, 16);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) yv[j] = bf16(0.f);
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) Ys[wg_swz(c0 + j, mloc)] = yv[j];
+      // gathered X chunk (padding rows and col overflow -> zeros)
+      long xrow = m < M ? fwd_gather(m, kh, kw, g) : -1;
+      bf16 xv[8];
+      if (xrow >= 0 && k0 + c0 + 8 <= Cin) {
+        uint4 raw = *(const uint4*)(X + xrow * (long)Cin + k0 + c0);
+        __builtin_memcpy(xv, &raw, 16);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) xv[j] = bf16(0.f);
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) Xs[wg_swz(c0 + j, mloc)] = xv[j];
     }
     __syncthreads();
 
-    const long mvalid = min((long)32, M - m0);
     bf16x8 a[4], b[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
-#pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        int m = fq * 8 + u;
-        bf16 v = m < mvalid ? Ys[m * 128 + wn + i * 16 + fr] : bf16(0.f);
-        short s;
-        __builtin_memcpy(&s, &v, 2);
-        a[i][u] = s;
-      }
+      a[i] = *(const bf16x8*)&Ys[wg_swz(wn + i * 16 + fr, fq * 8)];
 #pragma unroll
     for (int j = 0; j < 4; ++j)
-#pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        int m = fq * 8 + u;
-        bf16 v = m < mvalid ? Xs[m * 128 + wk + j * 16 + fr] : bf16(0.f);
-        short s;
-        __builtin_memcpy(&s, &v, 2);
-        b[j][u] = s;
-      }
+      b[j] = *(const bf16x8*)&Xs[wg_swz(wk + j * 16 + fr, fq * 8)];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
 #pragma unroll
@@ -413,13 +412,12 @@ at::Tensor conv3x3_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn, long H,
   long tiles = (long)nbn * nbk;
   int msplit = (int)std::max<long>(
       1, std::min<long>((M + 31) / 32, 1024 / tiles));
-  auto zp = zero_page_for(x2d);
   auto stream = at::cuda::getCurrentCUDAStream();
   for (int tap = 0; tap < 9; ++tap) {
     conv3x3_wgrad_kernel<<<(int)(tiles * msplit), GEMM_TPB, 0, stream>>>(
         (const bf16*)dy2d.const_data_ptr(), (const bf16*)x2d.const_data_ptr(),
         dW.data_ptr<float>(), M, (int)Cout, (int)Cin, g, tap, nbn, nbk,
-        msplit, (const bf16*)zp.const_data_ptr());
+        msplit);
     CHECK_CUDA_OK();
   }
   return dW;

@@ -141,16 +141,53 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
 }
 
 // ---- TN GEMM for wgrad: dW[N,K] += sum_m dY[m,n] * X[m,k] ----------------
-// Tiles: output 128n x 128k, M chunked by 32 per K-step and split across
-// blocks (split-M) with fp32 atomic accumulation into dW.
-// LDS chunks are [32m][128c]; fragments need the m-major (transposed) read,
-// which is a strided ds_read per element (v1; tr_b16 upgrade is planned).
+// Output tiles 128n x 128k, M chunked by 32 and split across blocks
+// (split-M) with fp32 atomic accumulation into dW.
+//
+// The MFMA fragments need the m-major (transposed) view of both chunks, so
+// staging is global->reg->TRANSPOSED LDS write ([col][32 m] layout with an
+// 8-element-block XOR swizzle to spread write banks); fragment loads are
+// then contiguous ds_read_b128.  Out-of-range rows/cols stage zeros, which
+// also removes per-fragment masking.
+__device__ __forceinline__ int tn_swz(int c, int m) {
+  return c * 32 + (m ^ (((c >> 3) & 3) << 3));
+}
+
+// stage a [32 m][128 col] chunk of a [M x ld] matrix, transposed, into LDS
+__device__ __forceinline__ void tn_stage(const bf16* __restrict__ g, long ld,
+                                         long m0, long M, long col0,
+                                         bf16* lds) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int rnd = 0; rnd < 2; ++rnd) {
+    const int unit = rnd * GEMM_TPB + t;  // 16 units (of 8 cols) per m-row
+    const long m = m0 + (unit >> 4);
+    const int c0 = (unit & 15) * 8;
+    bf16 vals[8];
+    if (m < M && col0 + c0 + 8 <= ld) {
+      uint4 raw = *(const uint4*)(g + m * ld + col0 + c0);
+      __builtin_memcpy(vals, &raw, 16);
+    } else if (m < M && col0 + c0 < ld) {  // ragged col tail
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        vals[j] = (col0 + c0 + j < ld) ? g[m * ld + col0 + c0 + j]
+                                       : bf16(0.f);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vals[j] = bf16(0.f);
+    }
+    const int mloc = unit >> 4;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) lds[tn_swz(c0 + j, mloc)] = vals[j];
+  }
+}
+
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 gemm_tn_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
                float* __restrict__ dW, long M, long N, long K, int nbn,
-               int nbk, int msplit, const bf16* __restrict__ zero_page) {
-  __shared__ bf16 Ys[32 * 128];
-  __shared__ bf16 Xs[32 * 128];
+               int nbk, int msplit) {
+  __shared__ bf16 Ys[128 * 32];
+  __shared__ bf16 Xs[128 * 32];
 
   const int tiles = nbn * nbk;
   const int tile = blockIdx.x % tiles;
@@ -180,55 +217,18 @@ gemm_tn_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
   for (long mc = mc0; mc < mc1; ++mc) {
     const long m0 = mc * 32;
     __syncthreads();
-    // stage [32][128] chunks: 32 rows x 256B = 8 KiB each
-    {
-      const int tt = threadIdx.x;
-#pragma unroll
-      for (int rnd = 0; rnd < 2; ++rnd) {
-        int unit = rnd * GEMM_TPB + tt;  // 0..511, 16 units per row
-        long m = m0 + (unit >> 4);
-        if (m >= M) m = M - 1;
-        int coff = (unit & 15) * 8;
-        // rows can be narrower than the 128-col chunk (N or K < 128):
-        // out-of-row columns read the zero page instead of faulting
-        const bf16* yp = (n0 + coff + 8 <= N) ? dY + m * N + n0 + coff
-                                              : zero_page;
-        const bf16* xp = (k0 + coff + 8 <= K) ? X + m * K + k0 + coff
-                                              : zero_page;
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int*)yp,
-            (__attribute__((address_space(3))) unsigned int*)(Ys + unit * 8),
-            16, 0, 0);
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int*)xp,
-            (__attribute__((address_space(3))) unsigned int*)(Xs + unit * 8),
-            16, 0, 0);
-      }
-    }
+    tn_stage(dY, N, m0, M, n0, Ys);
+    tn_stage(X, K, m0, M, k0, Xs);
     __syncthreads();
 
-    const long mvalid = min((long)32, M - m0);
-    // A fragment: dY^T[n][m]: lane holds dY[m=fq*8+u][n0+wn+i*16+fr]
-    // B fragment: X[m=fq*8+u][k0+wk+j*16+fr]
+    // A fragment: dY^T[n][m]; B fragment: X^T -> both contiguous b128 reads
     bf16x8 a[4], b[4];
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < 4; ++i)
+      a[i] = *(const bf16x8*)&Ys[tn_swz(wn + i * 16 + fr, fq * 8)];
 #pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        int m = fq * 8 + u;
-        a[i][u] = bf16_bits(
-            m < mvalid ? Ys[m * 128 + wn + i * 16 + fr] : bf16(0.f));
-      }
-    }
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-#pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        int m = fq * 8 + u;
-        b[j][u] = bf16_bits(
-            m < mvalid ? Xs[m * 128 + wk + j * 16 + fr] : bf16(0.f));
-      }
-    }
+    for (int j = 0; j < 4; ++j)
+      b[j] = *(const bf16x8*)&Xs[tn_swz(wk + j * 16 + fr, fq * 8)];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
 #pragma unroll
@@ -308,6 +308,7 @@ at::Tensor gemm_tn(at::Tensor dY, at::Tensor X, long msplit) {
   auto Xc = X.contiguous();
   long M = Yc.size(0), N = Yc.size(1), K = Xc.size(1);
   TORCH_CHECK(Xc.size(0) == M, "M mismatch");
+  TORCH_CHECK(N % 8 == 0 && K % 8 == 0, "N,K must be multiples of 8");
   auto dW = at::zeros({N, K}, Yc.options().dtype(at::kFloat));
   int nbn = (int)((N + 127) / 128), nbk = (int)((K + 127) / 128);
   if (msplit <= 0) {
@@ -315,12 +316,10 @@ at::Tensor gemm_tn(at::Tensor dY, at::Tensor X, long msplit) {
     long tiles = (long)nbn * nbk;
     msplit = std::max<long>(1, std::min<long>((M + 31) / 32, 1024 / tiles));
   }
-  auto zp = at::zeros({16}, Yc.options());
   auto stream = at::cuda::getCurrentCUDAStream();
   gemm_tn_kernel<<<(int)(nbn * nbk * msplit), GEMM_TPB, 0, stream>>>(
       (const bf16*)Yc.const_data_ptr(), (const bf16*)Xc.const_data_ptr(),
-      dW.data_ptr<float>(), M, N, K, nbn, nbk, (int)msplit,
-      (const bf16*)zp.const_data_ptr());
+      dW.data_ptr<float>(), M, N, K, nbn, nbk, (int)msplit);
   CHECK_CUDA_OK();
   return dW;
 }

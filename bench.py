@@ -96,9 +96,11 @@ def main():
             out = model(x)
         loss = criterion(out, targets)
         if hasattr(model, "reducer"):
-            model.zero_grad()
+            model.zero_grad()  # zero the bucket flats (grads are views)
         else:
-            optimizer.zero_grad(set_to_none=False)
+            # fresh grads each step: avoids 161 fill_ kernels + 161
+            # AccumulateGrad add_ kernels per step
+            optimizer.zero_grad(set_to_none=True)
         loss.backward()
         optimizer.step()
         return loss
