@@ -121,6 +121,52 @@ def conv3x3_mfma(x: torch.Tensor, weight: torch.Tensor,
     return _Conv3x3.apply(x, weight, stride)
 
 
+class _ConvGeneric(torch.autograd.Function):
+    """Generic implicit-GEMM conv (element-gather im2col) — serves the 7x7
+    stem.  Input gradient is not implemented (the stem input is data); the
+    dispatcher guards on ``x.requires_grad``."""
+
+    @staticmethod
+    @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
+    def forward(ctx, x: torch.Tensor, weight: torch.Tensor, stride: int,
+                pad: int):
+        e = require_ext()
+        n, cin, h, w = x.shape
+        cout, _, kh, kw = weight.shape
+        xc = x.contiguous(memory_format=torch.channels_last)
+        x2d = _rows(xc)
+        k = kh * kw * cin
+        kpad = (k + 31) // 32 * 32
+        w2 = weight.contiguous(memory_format=torch.channels_last) \
+            .permute(0, 2, 3, 1).reshape(cout, k)
+        if kpad != k:
+            w2 = torch.nn.functional.pad(w2, (0, kpad - k))
+        y2d = e.conv_generic_fwd(x2d, n, h, w, kh, kw, stride, pad, w2)
+        ctx.save_for_backward(x2d)
+        ctx.meta = (n, cin, h, w, kh, kw, stride, pad, cout, k)
+        ho = (h + 2 * pad - kh) // stride + 1
+        wo = (w + 2 * pad - kw) // stride + 1
+        return y2d.view(n, ho, wo, cout).permute(0, 3, 1, 2)
+
+    @staticmethod
+    @torch.amp.custom_bwd(device_type="cuda")
+    def backward(ctx, grad_y: torch.Tensor):
+        e = require_ext()
+        (x2d,) = ctx.saved_tensors
+        n, cin, h, w, kh, kw, stride, pad, cout, k = ctx.meta
+        gy2d = _rows(grad_y.contiguous(memory_format=torch.channels_last)) \
+            .to(torch.bfloat16)
+        dw2 = e.conv_generic_wgrad(gy2d, x2d, n, h, w, kh, kw, stride, pad)
+        dw = dw2[:, :k].view(cout, kh, kw, cin).permute(0, 3, 1, 2) \
+            .contiguous(memory_format=torch.channels_last)
+        return None, dw, None, None
+
+
+def conv_stem_mfma(x: torch.Tensor, weight: torch.Tensor, stride: int,
+                   pad: int) -> torch.Tensor:
+    return _ConvGeneric.apply(x, weight, stride, pad)
+
+
 def _conv1x1_env_default() -> str:
     return os.environ.get("AMDTRAIN_CONV1X1", "custom")
 
@@ -138,15 +184,25 @@ class AmdConv2d(nn.Conv2d):
         # fp32 inference/training falls through to MIOpen
         bf16_ok = (x.dtype == torch.bfloat16
                    or torch.is_autocast_enabled("cuda"))
-        if (x.is_cuda and bf16_ok and ext_available() and self.bias is None
-                and self.in_channels % 32 == 0
-                and self.out_channels % 16 == 0):
-            if (self.kernel_size == (1, 1) and self.padding == (0, 0)
+        if x.is_cuda and bf16_ok and ext_available() and self.bias is None:
+            ch_ok = (self.in_channels % 32 == 0
+                     and self.out_channels % 16 == 0)
+            if (ch_ok and self.kernel_size == (1, 1)
+                    and self.padding == (0, 0)
                     and _conv1x1_env_default() == "custom"):
                 return conv1x1_mfma(x, self.weight, self.stride[0])
-            if (self.kernel_size == (3, 3) and self.padding == (1, 1)
+            if (ch_ok and self.kernel_size == (3, 3)
+                    and self.padding == (1, 1)
                     and self.stride[0] in (1, 2)
                     and os.environ.get("AMDTRAIN_CONV3X3", "custom")
                     == "custom"):
                 return conv3x3_mfma(x, self.weight, self.stride[0])
+            if (not x.requires_grad and self.out_channels % 16 == 0
+                    and self.stride[0] == self.stride[1]
+                    and self.padding[0] == self.padding[1]
+                    and os.environ.get("AMDTRAIN_CONVSTEM", "custom")
+                    == "custom"):
+                # generic element-gather path (the 7x7 stem; no input grad)
+                return conv_stem_mfma(x, self.weight, self.stride[0],
+                                      self.padding[0])
         return super().forward(x)

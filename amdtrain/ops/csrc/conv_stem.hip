@@ -1,0 +1,285 @@
+// Generic implicit-GEMM convolution for gfx950 — covers the ResNet stem
+// (7x7 stride-2 pad-3, Cin=3) and any other odd configuration.
+//
+// Motivation (measured): torch-ROCm's MIOpen path for the stem wgrad
+// nondeterministically selects `naive_conv_..._wrw` at ~1.2 s/call
+// (profiles/rocprof_kernel_stats_allcustom.txt) — the hot path must not
+// depend on MIOpen find.  Here the im2col K dimension (KH*KW*Cin, e.g. 147)
+// is padded to a multiple of 32 and A-tiles are staged by per-element
+// gather (scalar loads + LDS writes; the data is tiny — Cin=3 — so staging
+// amplification is ~940 MB/step at b256, ~0.15 ms of HBM time), then the
+// same 128x128 MFMA block structure as gemm.hip runs the math.
+//
+//   fwd:   Y[m, cout] = sum_col im2col[m, col] * W2[cout, col]
+//   wgrad: dW2[cout, col] = sum_m dY[m, cout] * im2col[m, col]
+// (input dgrad is not needed: the stem input is data.)
+#include "common.h"
+
+namespace {
+
+using bf16 = __hip_bfloat16;
+using bf16x8 = __attribute__((ext_vector_type(8))) short;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+constexpr int GEMM_TPB = 256;
+constexpr int BK = 32;
+
+struct StemGeom {
+  int H, W, Hout, Wout;    // input / output spatial
+  int KH, KW, Cin;         // filter
+  int stride, pad;
+  int Kpad;                // padded im2col K (multiple of 32)
+};
+
+// element (m, col) of the im2col matrix -> value index in x, or -1
+__device__ __forceinline__ long im2col_index(long m, int col,
+                                             const StemGeom& g) {
+  if (col >= g.KH * g.KW * g.Cin) return -1;  // K padding
+  long t = m;
+  const int wo = (int)(t % g.Wout); t /= g.Wout;
+  const int ho = (int)(t % g.Hout); t /= g.Hout;
+  const int ci = col % g.Cin;
+  const int tap = col / g.Cin;
+  const int kh = tap / g.KW, kw = tap % g.KW;
+  const int h = ho * g.stride - g.pad + kh;
+  const int w = wo * g.stride - g.pad + kw;
+  if (h < 0 || h >= g.H || w < 0 || w >= g.W) return -1;
+  return ((t * g.H + h) * (long)g.W + w) * g.Cin + ci;
+}
+
+// stage a [128 m][32 col] im2col tile into LDS (per-element gather)
+__device__ __forceinline__ void stage_im2col(
+    const bf16* __restrict__ x, long m0, long M, int c0, const StemGeom& g,
+    bf16* lds) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int rnd = 0; rnd < 2; ++rnd) {
+    const int unit = rnd * GEMM_TPB + t;  // 4 units of 8 cols per m-row
+    long m = m0 + (unit >> 2);
+    const bool valid_m = m < M;
+    if (!valid_m) m = M - 1;
+    const int cc0 = c0 + (unit & 3) * 8;
+    bf16 v[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      long idx = valid_m ? im2col_index(m, cc0 + j, g) : -1;
+      v[j] = idx < 0 ? bf16(0.f) : x[idx];
+    }
+    *(Pack<bf16, 8>*)(lds + unit * 8) = *(Pack<bf16, 8>*)v;
+  }
+}
+
+__device__ __forceinline__ void stage_rows(
+    const bf16* __restrict__ gsrc, long ld, long row0, long rows, long koff,
+    bf16* lds) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int rnd = 0; rnd < 2; ++rnd) {
+    int unit = rnd * GEMM_TPB + t;
+    long row = row0 + (unit >> 2);
+    if (row >= rows) row = rows - 1;
+    const bf16* src = gsrc + row * ld + koff + (unit & 3) * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)(lds + unit * 8), 16,
+        0, 0);
+  }
+}
+
+__global__ void __launch_bounds__(GEMM_TPB, 2)
+conv_generic_fwd_kernel(const bf16* __restrict__ x,
+                        const bf16* __restrict__ W2, bf16* __restrict__ Y,
+                        long M, int Cout, StemGeom g, int nbm, int nbn) {
+  __shared__ bf16 As[128 * BK];
+  __shared__ bf16 Bs[128 * BK];
+  const int bid = blockIdx.x;
+  const int bm = bid / nbn, bn = bid % nbn;
+  const long m0 = (long)bm * 128, n0 = (long)bn * 128;
+  const int t = threadIdx.x;
+  const int wave = t / AMD_WAVE, lane = t % AMD_WAVE;
+  const int wm = (wave >> 1) * 64, wn = (wave & 1) * 64;
+  const int fr = lane & 15, fq = lane >> 4;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int ks = 0; ks < g.Kpad / BK; ++ks) {
+    __syncthreads();
+    stage_im2col(x, m0, M, ks * BK, g, As);
+    stage_rows(W2, g.Kpad, n0, Cout, ks * BK, Bs);
+    __syncthreads();
+    bf16x8 a[4], b[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      a[i] = *(const bf16x8*)&As[(wm + i * 16 + fr) * BK + fq * 8];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      b[j] = *(const bf16x8*)&Bs[(wn + j * 16 + fr) * BK + fq * 8];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a[i], b[j], acc[i][j], 0, 0, 0);
+  }
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = m0 + wm + i * 16 + fq * 4 + r;
+        long col = n0 + wn + j * 16 + fr;
+        if (row < M && col < Cout)
+          Y[row * Cout + col] = __float2bfloat16(acc[i][j][r]);
+      }
+}
+
+// wgrad: transposed reg-staging like gemm.hip's gemm_tn, with the B chunk
+// gathered from im2col
+__device__ __forceinline__ int st_swz(int c, int m) {
+  return c * 32 + (m ^ (((c >> 3) & 3) << 3));
+}
+
+__global__ void __launch_bounds__(GEMM_TPB, 2)
+conv_generic_wgrad_kernel(const bf16* __restrict__ dY,
+                          const bf16* __restrict__ x, float* __restrict__ dW2,
+                          long M, int Cout, StemGeom g, int nbn, int nbk,
+                          int msplit) {
+  __shared__ bf16 Ys[128 * 32];
+  __shared__ bf16 Xs[128 * 32];
+  const int tiles = nbn * nbk;
+  const int tile = blockIdx.x % tiles;
+  const int mpart = blockIdx.x / tiles;
+  const int bn = tile / nbk, bk = tile % nbk;
+  const long n0 = (long)bn * 128, k0 = (long)bk * 128;
+
+  const long mchunks = (M + 31) / 32;
+  const long cpp = (mchunks + msplit - 1) / msplit;
+  const long mc0 = (long)mpart * cpp;
+  const long mc1 = min(mc0 + cpp, mchunks);
+
+  const int t = threadIdx.x;
+  const int wave = t / AMD_WAVE, lane = t % AMD_WAVE;
+  const int wn = (wave >> 1) * 64, wk = (wave & 1) * 64;
+  const int fr = lane & 15, fq = lane >> 4;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (long mc = mc0; mc < mc1; ++mc) {
+    const long m0 = mc * 32;
+    __syncthreads();
+#pragma unroll
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      const int unit = rnd * GEMM_TPB + t;
+      const long m = m0 + (unit >> 4);
+      const int c0 = (unit & 15) * 8;
+      const int mloc = unit >> 4;
+      bf16 yv[8];
+      if (m < M && n0 + c0 + 8 <= Cout) {
+        uint4 raw = *(const uint4*)(dY + m * Cout + n0 + c0);
+        __builtin_memcpy(yv, &raw, 16);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) yv[j] = bf16(0.f);
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) Ys[st_swz(c0 + j, mloc)] = yv[j];
+      bf16 xv[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        long idx = m < M ? im2col_index(m, (int)(k0 + c0 + j), g) : -1;
+        xv[j] = idx < 0 ? bf16(0.f) : x[idx];
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) Xs[st_swz(c0 + j, mloc)] = xv[j];
+    }
+    __syncthreads();
+
+    bf16x8 a[4], b[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      a[i] = *(const bf16x8*)&Ys[st_swz(wn + i * 16 + fr, fq * 8)];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      b[j] = *(const bf16x8*)&Xs[st_swz(wk + j * 16 + fr, fq * 8)];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a[i], b[j], acc[i][j], 0, 0, 0);
+  }
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long n = n0 + wn + i * 16 + fq * 4 + r;
+        long k = k0 + wk + j * 16 + fr;
+        if (n < Cout && k < g.Kpad)
+          atomicAdd(&dW2[n * (long)g.Kpad + k], acc[i][j][r]);
+      }
+}
+
+static StemGeom make_geom(long H, long W, long KH, long KW, long Cin,
+                          long stride, long pad) {
+  StemGeom g;
+  g.H = (int)H; g.W = (int)W;
+  g.Hout = (int)((H + 2 * pad - KH) / stride + 1);
+  g.Wout = (int)((W + 2 * pad - KW) / stride + 1);
+  g.KH = (int)KH; g.KW = (int)KW; g.Cin = (int)Cin;
+  g.stride = (int)stride; g.pad = (int)pad;
+  int K = (int)(KH * KW * Cin);
+  g.Kpad = (K + 31) / 32 * 32;
+  return g;
+}
+
+}  // namespace
+
+// x2d: [N*H*W, Cin] bf16 NHWC rows; w2: [Cout, Kpad] (host-padded)
+at::Tensor conv_generic_fwd(at::Tensor x2d, long Nn, long H, long W,
+                            long KH, long KW, long stride, long pad,
+                            at::Tensor w2) {
+  TORCH_CHECK(x2d.is_cuda() && x2d.scalar_type() == at::kBFloat16);
+  long Cin = x2d.size(1), Cout = w2.size(0);
+  auto g = make_geom(H, W, KH, KW, Cin, stride, pad);
+  TORCH_CHECK(w2.size(1) == g.Kpad, "weight must be K-padded to ", g.Kpad);
+  long M = Nn * g.Hout * g.Wout;
+  auto y = at::empty({M, Cout}, x2d.options());
+  int nbm = (int)((M + 127) / 128), nbn = (int)((Cout + 127) / 128);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  conv_generic_fwd_kernel<<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+      (const bf16*)x2d.const_data_ptr(), (const bf16*)w2.const_data_ptr(),
+      (bf16*)y.data_ptr(), M, (int)Cout, g, nbm, nbn);
+  CHECK_CUDA_OK();
+  return y;
+}
+
+at::Tensor conv_generic_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn,
+                              long H, long W, long KH, long KW, long stride,
+                              long pad) {
+  long Cin = x2d.size(1), Cout = dy2d.size(1);
+  auto g = make_geom(H, W, KH, KW, Cin, stride, pad);
+  long M = Nn * g.Hout * g.Wout;
+  TORCH_CHECK(dy2d.size(0) == M);
+  auto dW2 = at::zeros({Cout, g.Kpad}, dy2d.options().dtype(at::kFloat));
+  int nbn = (int)((Cout + 127) / 128), nbk = (int)((g.Kpad + 127) / 128);
+  long tiles = (long)nbn * nbk;
+  int msplit = (int)std::max<long>(
+      1, std::min<long>((M + 31) / 32, 2048 / tiles));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  conv_generic_wgrad_kernel<<<(int)(tiles * msplit), GEMM_TPB, 0, stream>>>(
+      (const bf16*)dy2d.const_data_ptr(), (const bf16*)x2d.const_data_ptr(),
+      dW2.data_ptr<float>(), M, (int)Cout, g, nbn, nbk, msplit);
+  CHECK_CUDA_OK();
+  return dW2;
+}

@@ -52,6 +52,13 @@ at::Tensor conv3x3_dgrad(at::Tensor dy2d, long Nn, long H, long W,
 at::Tensor conv3x3_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn, long H,
                          long W, long stride);
 
+// conv_stem.hip
+at::Tensor conv_generic_fwd(at::Tensor x2d, long Nn, long H, long W, long KH,
+                            long KW, long stride, long pad, at::Tensor w2);
+at::Tensor conv_generic_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn,
+                              long H, long W, long KH, long KW, long stride,
+                              long pad);
+
 // pool.hip
 std::vector<at::Tensor> max_pool_3x3_s2_fwd(at::Tensor x);
 at::Tensor max_pool_3x3_s2_bwd(at::Tensor grad_y, at::Tensor idx, long H,
@@ -86,6 +93,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_fwd", &conv3x3_fwd);
   m.def("conv3x3_dgrad", &conv3x3_dgrad);
   m.def("conv3x3_wgrad", &conv3x3_wgrad);
+  m.def("conv_generic_fwd", &conv_generic_fwd);
+  m.def("conv_generic_wgrad", &conv_generic_wgrad);
   m.def("max_pool_3x3_s2_fwd", &max_pool_3x3_s2_fwd);
   m.def("max_pool_3x3_s2_bwd", &max_pool_3x3_s2_bwd);
   m.def("global_avg_pool_fwd", &global_avg_pool_fwd);

@@ -90,3 +90,31 @@ def test_resnet18_output_parity_custom_vs_miopen(monkeypatch):
         y_miopen = m(x)
     assert torch.allclose(y_custom.float(), y_miopen.float(), atol=0.5,
                           rtol=0.05), (y_custom - y_miopen).abs().max()
+
+
+@pytest.mark.parametrize("cfg", [
+    (3, 64, 7, 2, 3, 112),   # the ResNet stem
+    (32, 64, 5, 1, 2, 17),   # odd generic config
+])
+def test_conv_generic_stem(cfg):
+    from amdtrain.ops.conv import conv_stem_mfma
+    _ext()
+    cin, cout, k, stride, pad, hw = cfg
+    torch.manual_seed(5)
+    x = torch.randn(2, cin, hw, hw, device=DEV) \
+        .contiguous(memory_format=torch.channels_last)
+    w = torch.randn(cout, cin, k, k, device=DEV) * ((k * k * cin) ** -0.5)
+    wg = w.clone().requires_grad_(True)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        y = conv_stem_mfma(x, wg, stride, pad)
+    g = torch.randn_like(y.float())
+    y.float().backward(g)
+
+    wr = w.clone().float().requires_grad_(True)
+    yr = F.conv2d(x.float(), wr, stride=stride, padding=pad)
+    yr.backward(g)
+    assert y.shape == yr.shape
+    assert torch.allclose(y.float(), yr, atol=0.3, rtol=0.05), \
+        (y.float() - yr).abs().max()
+    assert torch.allclose(wg.grad.float(), wr.grad, atol=2.0, rtol=0.1), \
+        (wg.grad.float() - wr.grad).abs().max()
