@@ -1,4 +1,7 @@
 """Localize conv3x3 forward mismatches: error map by position, per-tap checks."""
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 import torch.nn.functional as F
 
