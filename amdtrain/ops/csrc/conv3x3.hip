@@ -324,8 +324,9 @@ conv3x3_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
       }
 }
 
-// permute weight [Cout, 9, Cin] -> rotated [Cin, 9, Cout] for dgrad:
-// W'[cin][tap][cout] = W[cout][8 - tap][cin]
+// permute weight [Cout, 9, Cin] -> [Cin, 9, Cout] for dgrad.
+// NO 180-degree rotation here: the dgrad gather (ho = h+1-kh) already
+// encodes it, so tap k pairs with W[., k, .] directly.
 __global__ void __launch_bounds__(AMD_TPB)
 rotate_weight_kernel(const bf16* __restrict__ w, bf16* __restrict__ out,
                      int Cout, int Cin) {
@@ -337,7 +338,7 @@ rotate_weight_kernel(const bf16* __restrict__ w, bf16* __restrict__ out,
     const int tap = (int)(t % 9); t /= 9;
     const int cout = (int)t;
     out[((long)cin * 9 + tap) * Cout + cout] =
-        w[((long)cout * 9 + (8 - tap)) * Cin + cin];
+        w[((long)cout * 9 + tap) * Cin + cin];
   }
 }
 

@@ -46,6 +46,9 @@ def parse_args():
 
 def main():
     args = parse_args()
+    # MIOpen exhaustive find can stall for minutes on new shapes (observed at
+    # b512); FAST find keeps the library paths predictable for A/B runs
+    os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
     if args.no_ext:
         os.environ["AMDTRAIN_DISABLE_EXT"] = "1"
         os.environ["AMDTRAIN_ALLOW_EAGER"] = "1"
