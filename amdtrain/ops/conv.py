@@ -60,14 +60,11 @@ class _Conv1x1(torch.autograd.Function):
         # dgrad: dX = dY x W  (BT form with pre-transposed weight)
         wT = e.transpose_2d(w2d)                      # [Cin, Cout]
         dx2d = e.gemm_bt(gy2d, wT, False)             # [R_sub, Cin] bf16
-        dxs = dx2d.view(n, ho, wo, cin).permute(0, 3, 1, 2)
         if stride > 1:
-            dx = torch.zeros(n, cin, h, w, dtype=dxs.dtype,
-                             device=dxs.device) \
-                .contiguous(memory_format=torch.channels_last)
-            dx[:, :, ::stride, ::stride] = dxs
+            # fused zero+scatter (one write pass; stride-2 only in ResNet)
+            dx = e.scatter_rows_x2(dx2d, n, h, w, ho, wo)
         else:
-            dx = dxs
+            dx = dx2d.view(n, ho, wo, cin).permute(0, 3, 1, 2)
         # wgrad: dW[Cout,Cin] = dY^T x X (fp32 split-M accumulation)
         dw = e.gemm_tn(gy2d, x2d, 0).reshape(cout, cin, 1, 1)
         return dx, dw, None

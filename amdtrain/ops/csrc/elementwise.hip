@@ -110,7 +110,60 @@ mt_cast_kernel(MTMeta meta) {
   }
 }
 
+// scatter_rows_x2: dx[n,h,w,:] = (h,w both even) ? src[n,h/2,w/2,:] : 0
+// (stride-2 1x1-conv dgrad epilogue — one write pass instead of a zero fill
+// plus a strided copy)
+template <typename T, int VEC>
+__global__ void __launch_bounds__(AMD_TPB)
+scatter_rows_x2_kernel(const T* __restrict__ src, T* __restrict__ dst,
+                       long N, int H, int W, int Hs, int Ws, int C) {
+  const int gpr = C / VEC;
+  const long total = N * H * W * gpr;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long t = i;
+    const int gc = (int)(t % gpr); t /= gpr;
+    const int w = (int)(t % W); t /= W;
+    const int h = (int)(t % H); t /= H;
+    const long n = t;
+    Pack<T, VEC> v;
+    if (((h | w) & 1) == 0 && h / 2 < Hs && w / 2 < Ws) {
+      v = *(const Pack<T, VEC>*)(
+          src + (((n * Hs + h / 2) * Ws) + w / 2) * (long)C + gc * VEC);
+    } else {
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) v.v[k] = from_f32<T>(0.f);
+    }
+    *(Pack<T, VEC>*)(dst + i * VEC) = v;
+  }
+}
+
 }  // namespace
+
+at::Tensor scatter_rows_x2(at::Tensor src2d, long Nn, long H, long W,
+                           long Hs, long Ws) {
+  // src2d: [Nn*Hs*Ws, C] -> [Nn, C, H, W] channels_last with zeros at odd
+  TORCH_CHECK(src2d.is_cuda() && src2d.dim() == 2);
+  long C = src2d.size(1);
+  auto out = at::empty({Nn, C, H, W},
+                       src2d.options().memory_format(
+                           at::MemoryFormat::ChannelsLast));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, src2d.scalar_type(),
+      "scatter_rows_x2", [&] {
+        using devT = typename DevT<scalar_t>::type;
+        constexpr int VEC = 16 / sizeof(devT);
+        TORCH_CHECK(C % VEC == 0);
+        long total = Nn * H * W * (C / VEC);
+        scatter_rows_x2_kernel<devT, VEC>
+            <<<amd_grid(total), AMD_TPB, 0, stream>>>(
+                (const devT*)src2d.const_data_ptr(), (devT*)out.data_ptr(),
+                Nn, (int)H, (int)W, (int)Hs, (int)Ws, (int)C);
+        CHECK_CUDA_OK();
+      });
+  return out;
+}
 
 at::Tensor normalize_u8(at::Tensor x, std::vector<double> mean,
                         std::vector<double> std_, bool bf16_out) {

@@ -20,6 +20,8 @@ at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor target,
 at::Tensor normalize_u8(at::Tensor x, std::vector<double> mean,
                         std::vector<double> std_, bool bf16_out);
 at::Tensor topk_ranks(at::Tensor logits, at::Tensor target);
+at::Tensor scatter_rows_x2(at::Tensor src2d, long Nn, long H, long W,
+                           long Hs, long Ws);
 void multi_tensor_scale_check(std::vector<at::Tensor> tensors, double scale,
                               at::Tensor found_inf);
 void multi_tensor_cast(std::vector<at::Tensor> src,
@@ -73,6 +75,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_bwd", &cross_entropy_bwd);
   m.def("normalize_u8", &normalize_u8);
   m.def("topk_ranks", &topk_ranks);
+  m.def("scatter_rows_x2", &scatter_rows_x2);
   m.def("multi_tensor_scale_check", &multi_tensor_scale_check);
   m.def("multi_tensor_cast", &multi_tensor_cast);
   m.def("batch_norm_fwd_train", &batch_norm_fwd_train,
