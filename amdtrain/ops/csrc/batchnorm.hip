@@ -447,7 +447,7 @@ at::Tensor batch_norm_fwd_eval(at::Tensor x, at::Tensor weight,
 std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
                                        at::Tensor y, at::Tensor weight,
                                        at::Tensor mean, at::Tensor invstd,
-                                       bool relu) {
+                                       bool relu, bool need_ghat) {
   check_nhwc(x);
   check_nhwc(grad_out);
   const long C = x.size(1);
@@ -462,7 +462,7 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
   auto Dc = at::empty({C}, opts);
   auto gx = at::empty_like(x);
   at::Tensor ghat;
-  if (relu) ghat = at::empty_like(grad_out);
+  if (relu && need_ghat) ghat = at::empty_like(grad_out);
   auto stream = at::cuda::getCurrentCUDAStream();
 
   dispatch_vec(x, [&](auto* tp, auto vec) {
@@ -502,14 +502,15 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
           (const devT*)x.const_data_ptr(),                                  \
           (const devT*)grad_out.const_data_ptr(),                           \
           (const devT*)y.const_data_ptr(), (devT*)gx.data_ptr(),            \
-          RELU_ ? (devT*)ghat.data_ptr() : nullptr,                         \
+          WG_ ? (devT*)ghat.data_ptr() : nullptr,                         \
           A.data_ptr<float>(), Bc.data_ptr<float>(),            \
           Dc.data_ptr<float>(), total_vec, (int)C)
-    if (relu) BAPPLY(true, true);
+    if (relu && need_ghat) BAPPLY(true, true);
+    else if (relu) BAPPLY(true, false);
     else BAPPLY(false, false);
 #undef BAPPLY
     CHECK_CUDA_OK();
   });
-  if (!relu) ghat = grad_out;  // ghat == grad_out when no ReLU was fused
+  if (!(relu && need_ghat)) ghat = grad_out;  // placeholder / identity
   return {gx, gw, gb, ghat};
 }

@@ -39,7 +39,7 @@ at::Tensor batch_norm_fwd_eval(at::Tensor x, at::Tensor weight,
 std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
                                        at::Tensor y, at::Tensor weight,
                                        at::Tensor mean, at::Tensor invstd,
-                                       bool relu);
+                                       bool relu, bool need_ghat);
 
 // gemm.hip
 at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out);

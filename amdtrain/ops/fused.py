@@ -43,7 +43,7 @@ class _BNFunction(torch.autograd.Function):
         e = require_ext()
         grad_x, grad_w, grad_b, ghat = e.batch_norm_bwd(
             x, grad_out.contiguous(memory_format=torch.channels_last),
-            y, weight, mean, invstd, ctx.relu)
+            y, weight, mean, invstd, ctx.relu, ctx.has_addend)
         grad_addend = ghat if ctx.has_addend else None
         return (grad_x, grad_w, grad_b, None, None, None, None, None,
                 grad_addend)
