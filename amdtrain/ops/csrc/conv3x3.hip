@@ -232,7 +232,7 @@ __device__ __forceinline__ int wg_swz(int cc, int m) {
 
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 conv3x3_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
-                     float* __restrict__ part, long M, int Cout, int Cin,
+                     float* __restrict__ dW, long M, int Cout, int Cin,
                      ConvGeom g, int tap, int nbn, int nbk, int msplit) {
   __shared__ bf16 Ys[128 * 32];
   __shared__ bf16 Xs[128 * 32];
@@ -310,15 +310,18 @@ conv3x3_wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
             a[i], b[j], acc[i][j], 0, 0, 0);
   }
 
-  float* out = part + (long)blockIdx.x * 16384;
 #pragma unroll
   for (int i = 0; i < 4; ++i)
 #pragma unroll
     for (int j = 0; j < 4; ++j)
 #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        out[(wn + i * 16 + fq * 4 + r) * 128 + wk + j * 16 + fr] =
-            acc[i][j][r];
+      for (int r = 0; r < 4; ++r) {
+        long n = n0 + wn + i * 16 + fq * 4 + r;   // cout
+        long k = k0 + wk + j * 16 + fr;           // cin
+        if (n < Cout && k < Cin)
+          atomicAdd(&dW[n * (long)(9 * Cin) + (long)tap * Cin + k],
+                    acc[i][j][r]);
+      }
 }
 
 // permute weight [Cout, 9, Cin] -> [Cin, 9, Cout] for dgrad.
@@ -404,25 +407,18 @@ at::Tensor conv3x3_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn, long H,
   long Hout = (H + 2 - 3) / stride + 1, Wout = (W + 2 - 3) / stride + 1;
   long M = Nn * Hout * Wout;
   TORCH_CHECK(dy2d.size(0) == M);
-  auto dW = at::empty({Cout, 9 * Cin}, dy2d.options().dtype(at::kFloat));
+  auto dW = at::zeros({Cout, 9 * Cin}, dy2d.options().dtype(at::kFloat));
   ConvGeom g{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
   int nbn = (int)((Cout + 127) / 128), nbk = (int)((Cin + 127) / 128);
   long tiles = (long)nbn * nbk;
   int msplit = (int)std::max<long>(
-      1, std::min<long>((M + 31) / 32, 1024 / tiles));
-  auto part = at::empty({tiles * msplit, 16384},
-                        dy2d.options().dtype(at::kFloat));
+      1, std::min<long>((M + 31) / 32, 512 / tiles));
   auto stream = at::cuda::getCurrentCUDAStream();
   for (int tap = 0; tap < 9; ++tap) {
     conv3x3_wgrad_kernel<<<(int)(tiles * msplit), GEMM_TPB, 0, stream>>>(
         (const bf16*)dy2d.const_data_ptr(), (const bf16*)x2d.const_data_ptr(),
-        part.data_ptr<float>(), M, (int)Cout, (int)Cin, g, tap, nbn, nbk,
+        dW.data_ptr<float>(), M, (int)Cout, (int)Cin, g, tap, nbn, nbk,
         msplit);
-    CHECK_CUDA_OK();
-    tn_reduce_partials_kernel<<<amd_grid(tiles * 16384), AMD_TPB, 0,
-                                stream>>>(
-        part.data_ptr<float>(), dW.data_ptr<float>(), (int)tiles,
-        (int)msplit, nbk, Cout, Cin, 9 * Cin, (long)tap * Cin);
     CHECK_CUDA_OK();
   }
   return dW;

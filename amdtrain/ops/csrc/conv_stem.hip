@@ -146,7 +146,7 @@ __device__ __forceinline__ int st_swz(int c, int m) {
 
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 conv_generic_wgrad_kernel(const bf16* __restrict__ dY,
-                          const bf16* __restrict__ x, float* __restrict__ part,
+                          const bf16* __restrict__ x, float* __restrict__ dW2,
                           long M, int Cout, StemGeom g, int nbn, int nbk,
                           int msplit) {
   __shared__ bf16 Ys[128 * 32];
@@ -217,15 +217,17 @@ conv_generic_wgrad_kernel(const bf16* __restrict__ dY,
         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a[i], b[j], acc[i][j], 0, 0, 0);
   }
-  float* out = part + (long)blockIdx.x * 16384;
 #pragma unroll
   for (int i = 0; i < 4; ++i)
 #pragma unroll
     for (int j = 0; j < 4; ++j)
 #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        out[(wn + i * 16 + fq * 4 + r) * 128 + wk + j * 16 + fr] =
-            acc[i][j][r];
+      for (int r = 0; r < 4; ++r) {
+        long n = n0 + wn + i * 16 + fq * 4 + r;
+        long k = k0 + wk + j * 16 + fr;
+        if (n < Cout && k < g.Kpad)
+          atomicAdd(&dW2[n * (long)g.Kpad + k], acc[i][j][r]);
+      }
 }
 
 static StemGeom make_geom(long H, long W, long KH, long KW, long Cin,
@@ -269,21 +271,15 @@ at::Tensor conv_generic_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn,
   auto g = make_geom(H, W, KH, KW, Cin, stride, pad);
   long M = Nn * g.Hout * g.Wout;
   TORCH_CHECK(dy2d.size(0) == M);
-  auto dW2 = at::empty({Cout, g.Kpad}, dy2d.options().dtype(at::kFloat));
+  auto dW2 = at::zeros({Cout, g.Kpad}, dy2d.options().dtype(at::kFloat));
   int nbn = (int)((Cout + 127) / 128), nbk = (int)((g.Kpad + 127) / 128);
   long tiles = (long)nbn * nbk;
   int msplit = (int)std::max<long>(
-      1, std::min<long>((M + 31) / 32, 2048 / tiles));
-  auto part = at::empty({tiles * msplit, 16384},
-                        dy2d.options().dtype(at::kFloat));
+      1, std::min<long>((M + 31) / 32, 512 / tiles));
   auto stream = at::cuda::getCurrentCUDAStream();
   conv_generic_wgrad_kernel<<<(int)(tiles * msplit), GEMM_TPB, 0, stream>>>(
       (const bf16*)dy2d.const_data_ptr(), (const bf16*)x2d.const_data_ptr(),
-      part.data_ptr<float>(), M, (int)Cout, g, nbn, nbk, msplit);
-  CHECK_CUDA_OK();
-  tn_reduce_partials_kernel<<<amd_grid(tiles * 16384), AMD_TPB, 0, stream>>>(
-      part.data_ptr<float>(), dW2.data_ptr<float>(), (int)tiles,
-      (int)msplit, nbk, Cout, g.Kpad, g.Kpad, 0);
+      dW2.data_ptr<float>(), M, (int)Cout, g, nbn, nbk, msplit);
   CHECK_CUDA_OK();
   return dW2;
 }

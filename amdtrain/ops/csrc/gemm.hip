@@ -184,7 +184,7 @@ __device__ __forceinline__ void tn_stage(const bf16* __restrict__ g, long ld,
 
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 gemm_tn_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
-               float* __restrict__ part, long M, long N, long K, int nbn,
+               float* __restrict__ dW, long M, long N, long K, int nbn,
                int nbk, int msplit) {
   __shared__ bf16 Ys[128 * 32];
   __shared__ bf16 Xs[128 * 32];
@@ -237,16 +237,18 @@ gemm_tn_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
             a[i], b[j], acc[i][j], 0, 0, 0);
   }
 
-  // write the block's 128x128 partial tile (summed by tn_reduce_partials)
-  float* out = part + (long)blockIdx.x * 16384;
+  // atomic split-M accumulation (measured cheaper than a partial-buffer
+  // second pass, whose traffic scales with block count)
 #pragma unroll
   for (int i = 0; i < 4; ++i)
 #pragma unroll
     for (int j = 0; j < 4; ++j)
 #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        out[(wn + i * 16 + fq * 4 + r) * 128 + wk + j * 16 + fr] =
-            acc[i][j][r];
+      for (int r = 0; r < 4; ++r) {
+        long n = n0 + wn + i * 16 + fq * 4 + r;
+        long k = k0 + wk + j * 16 + fr;
+        if (n < N && k < K) atomicAdd(&dW[n * K + k], acc[i][j][r]);
+      }
 }
 
 // ---- small bf16 2D transpose (per-step weight transpose for dgrad) -------
@@ -306,21 +308,15 @@ at::Tensor gemm_tn(at::Tensor dY, at::Tensor X, long msplit) {
   long M = Yc.size(0), N = Yc.size(1), K = Xc.size(1);
   TORCH_CHECK(Xc.size(0) == M, "M mismatch");
   TORCH_CHECK(N % 8 == 0 && K % 8 == 0, "N,K must be multiples of 8");
-  auto dW = at::empty({N, K}, Yc.options().dtype(at::kFloat));
+  auto dW = at::zeros({N, K}, Yc.options().dtype(at::kFloat));
   int nbn = (int)((N + 127) / 128), nbk = (int)((K + 127) / 128);
   long tiles = (long)nbn * nbk;
   if (msplit <= 0)
-    msplit = std::max<long>(1, std::min<long>((M + 31) / 32, 1024 / tiles));
-  auto part = at::empty({tiles * msplit, 16384},
-                        Yc.options().dtype(at::kFloat));
+    msplit = std::max<long>(1, std::min<long>((M + 31) / 32, 512 / tiles));
   auto stream = at::cuda::getCurrentCUDAStream();
   gemm_tn_kernel<<<(int)(tiles * msplit), GEMM_TPB, 0, stream>>>(
       (const bf16*)Yc.const_data_ptr(), (const bf16*)Xc.const_data_ptr(),
-      part.data_ptr<float>(), M, N, K, nbn, nbk, (int)msplit);
-  CHECK_CUDA_OK();
-  tn_reduce_partials_kernel<<<amd_grid(tiles * 16384), AMD_TPB, 0, stream>>>(
-      part.data_ptr<float>(), dW.data_ptr<float>(), (int)tiles,
-      (int)msplit, nbk, N, K, K, 0);
+      dW.data_ptr<float>(), M, N, K, nbn, nbk, (int)msplit);
   CHECK_CUDA_OK();
   return dW;
 }
