@@ -86,31 +86,6 @@ __device__ __forceinline__ float wave_reduce_max(float v) {
   return v;
 }
 
-// ---- split-M partial reduction for TN (wgrad) GEMMs ------------------------
-// Each TN block writes its 128x128 fp32 tile to part[blockIdx][16384]
-// (non-atomic); this kernel sums the msplit partials per tile and stores
-// into the output at dW[n * ld + col0 + k].  Replaces the atomicAdd flush
-// whose per-address contention scaled with msplit.
-static __global__ void __launch_bounds__(AMD_TPB)
-tn_reduce_partials_kernel(const float* __restrict__ part,
-                          float* __restrict__ dW, int tiles, int msplit,
-                          int nbk, long N, long K, long ld, long col0) {
-  const long total = (long)tiles * 16384;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += (long)gridDim.x * blockDim.x) {
-    const int tile = (int)(i >> 14);
-    const int idx = (int)(i & 16383);
-    const int bn = tile / nbk, bk = tile % nbk;
-    const long n = (long)bn * 128 + (idx >> 7);
-    const long k = (long)bk * 128 + (idx & 127);
-    if (n >= N || k >= K) continue;
-    float s = 0.f;
-    for (int p = 0; p < msplit; ++p)
-      s += part[((long)p * tiles + tile) * 16384 + idx];
-    dW[n * ld + col0 + k] = s;
-  }
-}
-
 // ---- multi-tensor-apply metadata (apex amp_C style, struct by value) -------
 
 constexpr int MT_TENSORS = 24;

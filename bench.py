@@ -41,6 +41,9 @@ def parse_args():
                    help="A/B: plain-PyTorch ops instead of the HIP kernels")
     p.add_argument("--profile", type=str, default="",
                    help="write a torch.profiler kernel table here (3 steps)")
+    p.add_argument("--style", type=str, default="ddp",
+                   choices=["ddp", "apex", "horovod"],
+                   help="launch-style variant to benchmark (BASELINE configs)")
     return p.parse_args()
 
 
@@ -76,11 +79,25 @@ def main():
 
     model = build_model(args.arch).to(device) \
         .to(memory_format=torch.channels_last)
-    if world > 1:
-        model = NativeDDP(model, bucket_cap_mb=args.bucket_mb)
     criterion = CrossEntropyLoss()
     optimizer = FusedSGD(model.parameters(), lr=0.1, momentum=0.9,
                          weight_decay=1e-4)
+    amp_handle = None
+    if args.style == "apex":
+        from amdtrain.parallel import amp as amp_mod
+        model, optimizer = amp_mod.initialize(model, optimizer,
+                                              opt_level="O2", dtype=dtype)
+        amp_handle = amp_mod
+    if args.style == "horovod":
+        import amdtrain.comm as C
+        from amdtrain.parallel import Compression, DistributedOptimizer
+        C.broadcast_module_state(model, src=0)
+        optimizer = DistributedOptimizer(optimizer,
+                                         model.named_parameters(),
+                                         compression=Compression.fp16,
+                                         fusion_mb=args.bucket_mb)
+    elif world > 1:
+        model = NativeDDP(model, bucket_cap_mb=args.bucket_mb)
     model.train()
 
     # synthetic data of the benchmark shape, resident on GPU (per-rank seed)
@@ -92,7 +109,7 @@ def main():
 
     def step():
         x = OF.normalize_u8(images_u8, dtype=dtype)
-        if dtype == torch.bfloat16:
+        if dtype == torch.bfloat16 and amp_handle is None:
             with torch.autocast("cuda", dtype=torch.bfloat16):
                 out = model(x)
         else:
@@ -100,11 +117,17 @@ def main():
         loss = criterion(out, targets)
         if hasattr(model, "reducer"):
             model.zero_grad()  # zero the bucket flats (grads are views)
+        elif hasattr(optimizer, "reducer"):
+            optimizer.zero_grad()
         else:
             # fresh grads each step: avoids 161 fill_ kernels + 161
             # AccumulateGrad add_ kernels per step
             optimizer.zero_grad(set_to_none=True)
-        loss.backward()
+        if amp_handle is not None:
+            with amp_handle.scale_loss(loss, optimizer) as scaled:
+                scaled.backward()
+        else:
+            loss.backward()
         optimizer.step()
         return loss
 
@@ -154,6 +177,7 @@ def main():
                 "image_size": S,
                 "seq_len": None,
                 "parallelism": f"dp{world}",
+                "style": args.style,
                 "sec_per_epoch_est": round(1_281_167 / ips, 1),
                 "hip_ext": not args.no_ext,
             },
