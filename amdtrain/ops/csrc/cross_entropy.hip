@@ -28,13 +28,16 @@ ce_fwd_kernel(const T* __restrict__ logits, const long* __restrict__ target,
     s = s * __expf(m - m2) + __expf(v - m2);
     m = m2;
   }
-  // wave reduce (max, then rescaled sums)
+  // wave reduce (max, then rescaled sums).  Lanes beyond C keep m = -inf;
+  // exp(-inf - (-inf)) is NaN, so -inf partials contribute exactly 0.
 #pragma unroll
   for (int off = AMD_WAVE / 2; off > 0; off >>= 1) {
     float mo = __shfl_down(m, off, AMD_WAVE);
     float so = __shfl_down(s, off, AMD_WAVE);
     float m2 = fmaxf(m, mo);
-    s = s * __expf(m - m2) + so * __expf(mo - m2);
+    float fa = (m == -INFINITY) ? 0.f : __expf(m - m2);
+    float fb = (mo == -INFINITY) ? 0.f : __expf(mo - m2);
+    s = s * fa + so * fb;
     m = m2;
   }
   if (lane == 0) {
