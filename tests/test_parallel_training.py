@@ -110,3 +110,31 @@ def _slurm_rank_math():
 
 def test_slurm_global_rank_math():
     assert _slurm_rank_math() == list(range(8))
+
+
+def _apex_style_fn(rank, world, steps):
+    """AMP O1 + NativeDDP + scale_loss on gloo (the apex entrypoint's path)."""
+    import torch.nn.functional as TF
+    from amdtrain.parallel import NativeDDP, amp
+    from amdtrain.ops import FusedSGD
+    model = _tiny_model(seed=rank)
+    opt = FusedSGD(model.parameters(), lr=0.05, momentum=0.9)
+    model, opt = amp.initialize(model, opt, opt_level="O1",
+                                dtype=torch.bfloat16)
+    ddp = NativeDDP(model, bucket_cap_mb=0.0001)
+    for step in range(steps):
+        torch.manual_seed(3000 + step * world + rank)
+        x = torch.randn(6, 8)
+        t = torch.randint(0, 4, (6,))
+        ddp.zero_grad()
+        loss = TF.cross_entropy(ddp(x), t)
+        with amp.scale_loss(loss, opt) as scaled:
+            scaled.backward()
+        opt.step()
+    return [p.detach().clone() for p in model.parameters()]
+
+
+def test_apex_style_ddp_lockstep():
+    res = run_distributed(_apex_style_fn, world=2, args=(3,))
+    for a, b in zip(res[0], res[1]):
+        assert torch.allclose(a, b, atol=1e-6)
