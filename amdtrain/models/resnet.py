@@ -29,7 +29,7 @@ import torch.nn as nn
 from ..ops import fused as OF
 
 
-from ..ops.conv import AmdConv2d
+from ..ops.conv import AmdConv2d  # noqa: F401 (stem + helpers)
 
 
 def conv3x3(in_planes: int, out_planes: int, stride: int = 1) -> nn.Conv2d:
@@ -115,7 +115,7 @@ class ResNet(nn.Module):
                  zero_init_residual: bool = False):
         super().__init__()
         self.inplanes = 64
-        self.conv1 = nn.Conv2d(3, 64, kernel_size=7, stride=2, padding=3,
+        self.conv1 = AmdConv2d(3, 64, kernel_size=7, stride=2, padding=3,
                                bias=False)
         self.bn1 = FusedBatchNorm2d(64)
         self.relu = nn.ReLU(inplace=True)

@@ -68,3 +68,12 @@ def test_train_mode_updates_stats():
     m(torch.randn(2, 3, 64, 64))
     assert not torch.equal(rm, m.bn1.running_mean)
     assert m.bn1.num_batches_tracked == nbt + 1
+
+
+def test_all_convs_are_amd_dispatch():
+    from amdtrain.ops.conv import AmdConv2d
+    m = build_model("resnet50")
+    convs = [mod for mod in m.modules() if isinstance(mod, torch.nn.Conv2d)]
+    assert len(convs) == 53
+    assert all(isinstance(c, AmdConv2d) for c in convs), \
+        "every conv (incl. the stem) must go through the dispatching Conv2d"
