@@ -41,15 +41,14 @@ def pmc_stats(root, out):
                 continue
             suf = kd[0][len("rocpd_kernel_dispatch_"):]
             try:
-                q = (f"SELECT ks.display_name, s.string, SUM(pe.value) "
+                q = (f"SELECT ks.display_name, pi.name, SUM(pe.value) "
                      f"FROM rocpd_pmc_event_{suf} pe "
                      f"JOIN rocpd_info_pmc_{suf} pi ON pe.pmc_id=pi.id "
-                     f"JOIN rocpd_string_{suf} s ON pi.name_id=s.id "
                      f"JOIN rocpd_kernel_dispatch_{suf} k "
                      f"  ON pe.event_id=k.event_id "
                      f"JOIN rocpd_info_kernel_symbol_{suf} ks "
                      f"  ON k.kernel_id=ks.id "
-                     f"GROUP BY ks.display_name, s.string")
+                     f"GROUP BY ks.display_name, pi.name")
                 agg = collections.defaultdict(dict)
                 for kname, cname, v in cur.execute(q):
                     agg[kname][cname] = v
