@@ -31,38 +31,62 @@ struct StemGeom {
   int Kpad;                // padded im2col K (multiple of 32)
 };
 
-// element (m, col) of the im2col matrix -> value index in x, or -1
-__device__ __forceinline__ long im2col_index(long m, int col,
-                                             const StemGeom& g) {
-  if (col >= g.KH * g.KW * g.Cin) return -1;  // K padding
+// per-row (m) decode, hoisted out of the per-element gather: output pixel ->
+// input-space base coordinates
+struct StemCoord {
+  long n_off;
+  int hb, wb;
+};
+
+__device__ __forceinline__ StemCoord stem_decode(long m, const StemGeom& g) {
+  StemCoord u;
   long t = m;
   const int wo = (int)(t % g.Wout); t /= g.Wout;
   const int ho = (int)(t % g.Hout); t /= g.Hout;
-  const int ci = col % g.Cin;
-  const int tap = col / g.Cin;
-  const int kh = tap / g.KW, kw = tap % g.KW;
-  const int h = ho * g.stride - g.pad + kh;
-  const int w = wo * g.stride - g.pad + kw;
-  if (h < 0 || h >= g.H || w < 0 || w >= g.W) return -1;
-  return ((t * g.H + h) * (long)g.W + w) * g.Cin + ci;
+  u.n_off = t * (long)g.H * g.W;
+  u.hb = ho * g.stride - g.pad;
+  u.wb = wo * g.stride - g.pad;
+  return u;
 }
 
-// stage a [128 m][32 col] im2col tile into LDS (per-element gather)
+// per-column decode via reciprocal multiply (integer div is ~20 cyc and was
+// executed per element per K-step before)
+__device__ __forceinline__ void col_decode(int col, const StemGeom& g,
+                                           float inv_cin, float inv_kw,
+                                           int& kh, int& kw, int& ci) {
+  const int tap = (int)((float)col * inv_cin + 1e-4f);
+  ci = col - tap * g.Cin;
+  kh = (int)((float)tap * inv_kw + 1e-4f);
+  kw = tap - kh * g.KW;
+}
+
+// element gather using hoisted row coords
+__device__ __forceinline__ long stem_elem(const StemCoord& u, int col,
+                                          const StemGeom& g, float inv_cin,
+                                          float inv_kw) {
+  if (col >= g.KH * g.KW * g.Cin) return -1;  // K padding
+  int kh, kw, ci;
+  col_decode(col, g, inv_cin, inv_kw, kh, kw, ci);
+  const int h = u.hb + kh, w = u.wb + kw;
+  if (h < 0 || h >= g.H || w < 0 || w >= g.W) return -1;
+  return (u.n_off + (long)h * g.W + w) * g.Cin + ci;
+}
+
+// stage a [128 m][32 col] im2col tile into LDS (per-element gather with
+// hoisted per-row decode — uc[] is precomputed once per kernel)
 __device__ __forceinline__ void stage_im2col(
-    const bf16* __restrict__ x, long m0, long M, int c0, const StemGeom& g,
-    bf16* lds) {
+    const bf16* __restrict__ x, const StemCoord* uc, const bool* valid,
+    int c0, const StemGeom& g, float inv_cin, float inv_kw, bf16* lds) {
   const int t = threadIdx.x;
 #pragma unroll
   for (int rnd = 0; rnd < 2; ++rnd) {
     const int unit = rnd * GEMM_TPB + t;  // 4 units of 8 cols per m-row
-    long m = m0 + (unit >> 2);
-    const bool valid_m = m < M;
-    if (!valid_m) m = M - 1;
     const int cc0 = c0 + (unit & 3) * 8;
     bf16 v[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      long idx = valid_m ? im2col_index(m, cc0 + j, g) : -1;
+      long idx = valid[rnd] ? stem_elem(uc[rnd], cc0 + j, g, inv_cin, inv_kw)
+                            : -1;
       v[j] = idx < 0 ? bf16(0.f) : x[idx];
     }
     *(Pack<bf16, 8>*)(lds + unit * 8) = *(Pack<bf16, 8>*)v;
@@ -106,9 +130,19 @@ conv_generic_fwd_kernel(const bf16* __restrict__ x,
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+  const float inv_cin = 1.f / g.Cin, inv_kw = 1.f / g.KW;
+  StemCoord uc[2];
+  bool valid[2];
+#pragma unroll
+  for (int rnd = 0; rnd < 2; ++rnd) {
+    long m = m0 + (((rnd * GEMM_TPB) + t) >> 2);
+    valid[rnd] = m < M;
+    uc[rnd] = stem_decode(valid[rnd] ? m : M - 1, g);
+  }
+
   for (int ks = 0; ks < g.Kpad / BK; ++ks) {
     __syncthreads();
-    stage_im2col(x, m0, M, ks * BK, g, As);
+    stage_im2col(x, uc, valid, ks * BK, g, inv_cin, inv_kw, As);
     stage_rows(W2, g.Kpad, n0, Cout, ks * BK, Bs);
     __syncthreads();
     bf16x8 a[4], b[4];
@@ -173,6 +207,24 @@ conv_generic_wgrad_kernel(const bf16* __restrict__ dY,
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+  // col -> (kh, kw, ci) is FIXED per thread across all m-chunks: hoist it
+  const float inv_cin = 1.f / g.Cin, inv_kw = 1.f / g.KW;
+  const int K = g.KH * g.KW * g.Cin;
+  int ckh[2][8], ckw[2][8], cci[2][8];
+  bool cok[2][8];
+#pragma unroll
+  for (int rnd = 0; rnd < 2; ++rnd) {
+    const int unit = rnd * GEMM_TPB + t;
+    const int c0 = (unit & 15) * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int col = (int)(k0 + c0 + j);
+      cok[rnd][j] = col < K;
+      col_decode(cok[rnd][j] ? col : 0, g, inv_cin, inv_kw, ckh[rnd][j],
+                 ckw[rnd][j], cci[rnd][j]);
+    }
+  }
+
   for (long mc = mc0; mc < mc1; ++mc) {
     const long m0 = mc * 32;
     __syncthreads();
@@ -193,10 +245,14 @@ conv_generic_wgrad_kernel(const bf16* __restrict__ dY,
 #pragma unroll
       for (int j = 0; j < 8; ++j) Ys[st_swz(c0 + j, mloc)] = yv[j];
       bf16 xv[8];
+      StemCoord u = stem_decode(m < M ? m : 0, g);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        long idx = m < M ? im2col_index(m, (int)(k0 + c0 + j), g) : -1;
-        xv[j] = idx < 0 ? bf16(0.f) : x[idx];
+        const int h = u.hb + ckh[rnd][j], w = u.wb + ckw[rnd][j];
+        const bool ok = (m < M) && cok[rnd][j] && h >= 0 && h < g.H &&
+                        w >= 0 && w < g.W;
+        long idx = (u.n_off + (long)h * g.W + w) * g.Cin + cci[rnd][j];
+        xv[j] = ok ? x[idx] : bf16(0.f);
       }
 #pragma unroll
       for (int j = 0; j < 8; ++j) Xs[st_swz(c0 + j, mloc)] = xv[j];
