@@ -150,7 +150,9 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
 // then contiguous ds_read_b128.  Out-of-range rows/cols stage zeros, which
 // also removes per-fragment masking.
 __device__ __forceinline__ int tn_swz(int c, int m) {
-  return c * 32 + (m ^ (((c >> 3) & 3) << 3));
+  // XOR c-bits 1-2 as well: spreads the stride-1 fragment reads over
+  // 8 banks (2-way, free) while keeping the stride-8 writes at 4-way
+  return c * 32 + (m ^ (((((c >> 3) & 3) ^ ((c >> 1) & 3))) << 3));
 }
 
 // stage a [32 m][128 col] chunk of a [M x ld] matrix, transposed, into LDS
