@@ -227,7 +227,7 @@ conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
 // Transposed reg-staging + contiguous b128 fragment reads (see gemm.hip
 // gemm_tn); gathered X rows that fall in the padding stage zeros.
 __device__ __forceinline__ int wg_swz(int cc, int m) {
-  return cc * 32 + (m ^ (((cc >> 3) & 3) << 3));
+  return cc * 32 + (m ^ (((((cc >> 3) & 3) ^ ((cc >> 1) & 3))) << 3));
 }
 
 __global__ void __launch_bounds__(GEMM_TPB, 2)

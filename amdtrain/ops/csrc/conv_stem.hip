@@ -175,7 +175,7 @@ conv_generic_fwd_kernel(const bf16* __restrict__ x,
 // wgrad: transposed reg-staging like gemm.hip's gemm_tn, with the B chunk
 // gathered from im2col
 __device__ __forceinline__ int st_swz(int c, int m) {
-  return c * 32 + (m ^ (((c >> 3) & 3) << 3));
+  return c * 32 + (m ^ (((((c >> 3) & 3) ^ ((c >> 1) & 3))) << 3));
 }
 
 __global__ void __launch_bounds__(GEMM_TPB, 2)
