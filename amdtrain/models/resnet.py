@@ -32,9 +32,10 @@ from ..ops import fused as OF
 from ..ops.conv import AmdConv2d  # noqa: F401 (stem + helpers)
 
 
-def conv3x3(in_planes: int, out_planes: int, stride: int = 1) -> nn.Conv2d:
+def conv3x3(in_planes: int, out_planes: int, stride: int = 1,
+            groups: int = 1) -> nn.Conv2d:
     return AmdConv2d(in_planes, out_planes, kernel_size=3, stride=stride,
-                     padding=1, bias=False)
+                     padding=1, groups=groups, bias=False)
 
 
 def conv1x1(in_planes: int, out_planes: int, stride: int = 1) -> nn.Conv2d:
@@ -87,13 +88,15 @@ class Bottleneck(nn.Module):
     expansion = 4
 
     def __init__(self, inplanes: int, planes: int, stride: int = 1,
-                 downsample: Optional[nn.Module] = None):
+                 downsample: Optional[nn.Module] = None, groups: int = 1,
+                 base_width: int = 64):
         super().__init__()
-        self.conv1 = conv1x1(inplanes, planes)
-        self.bn1 = FusedBatchNorm2d(planes)
-        self.conv2 = conv3x3(planes, planes, stride)
-        self.bn2 = FusedBatchNorm2d(planes)
-        self.conv3 = conv1x1(planes, planes * self.expansion)
+        width = int(planes * (base_width / 64.0)) * groups
+        self.conv1 = conv1x1(inplanes, width)
+        self.bn1 = FusedBatchNorm2d(width)
+        self.conv2 = conv3x3(width, width, stride, groups)
+        self.bn2 = FusedBatchNorm2d(width)
+        self.conv3 = conv1x1(width, planes * self.expansion)
         self.bn3 = FusedBatchNorm2d(planes * self.expansion)
         self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
@@ -112,8 +115,11 @@ class Bottleneck(nn.Module):
 class ResNet(nn.Module):
     def __init__(self, block: Type[Union[BasicBlock, Bottleneck]],
                  layers: List[int], num_classes: int = 1000,
-                 zero_init_residual: bool = False):
+                 zero_init_residual: bool = False, groups: int = 1,
+                 width_per_group: int = 64):
         super().__init__()
+        self.groups = groups
+        self.base_width = width_per_group
         self.inplanes = 64
         self.conv1 = AmdConv2d(3, 64, kernel_size=7, stride=2, padding=3,
                                bias=False)
@@ -149,10 +155,13 @@ class ResNet(nn.Module):
                 conv1x1(self.inplanes, planes * block.expansion, stride),
                 FusedBatchNorm2d(planes * block.expansion),
             )
-        layers = [block(self.inplanes, planes, stride, downsample)]
+        kw = {}
+        if block is Bottleneck:
+            kw = dict(groups=self.groups, base_width=self.base_width)
+        layers = [block(self.inplanes, planes, stride, downsample, **kw)]
         self.inplanes = planes * block.expansion
         for _ in range(1, blocks):
-            layers.append(block(self.inplanes, planes))
+            layers.append(block(self.inplanes, planes, **kw))
         return nn.Sequential(*layers)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
@@ -187,12 +196,24 @@ def resnet152(num_classes: int = 1000, **kw) -> ResNet:
     return ResNet(Bottleneck, [3, 8, 36, 3], num_classes=num_classes, **kw)
 
 
+def resnext50_32x4d(num_classes: int = 1000, **kw) -> ResNet:
+    return ResNet(Bottleneck, [3, 4, 6, 3], num_classes=num_classes,
+                  groups=32, width_per_group=4, **kw)
+
+
+def wide_resnet50_2(num_classes: int = 1000, **kw) -> ResNet:
+    return ResNet(Bottleneck, [3, 4, 6, 3], num_classes=num_classes,
+                  width_per_group=128, **kw)
+
+
 _REGISTRY: Dict[str, Callable[..., nn.Module]] = {
     "resnet18": resnet18,
     "resnet34": resnet34,
     "resnet50": resnet50,
     "resnet101": resnet101,
     "resnet152": resnet152,
+    "resnext50_32x4d": resnext50_32x4d,
+    "wide_resnet50_2": wide_resnet50_2,
 }
 
 

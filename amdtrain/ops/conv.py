@@ -181,7 +181,9 @@ class AmdConv2d(nn.Conv2d):
         # fp32 inference/training falls through to MIOpen
         bf16_ok = (x.dtype == torch.bfloat16
                    or torch.is_autocast_enabled("cuda"))
-        if x.is_cuda and bf16_ok and ext_available() and self.bias is None:
+        if (x.is_cuda and bf16_ok and ext_available()
+                and self.bias is None and self.groups == 1
+                and self.dilation == (1, 1)):
             ch_ok = (self.in_channels % 32 == 0
                      and self.out_channels % 16 == 0)
             if (ch_ok and self.kernel_size == (1, 1)

@@ -13,6 +13,8 @@ PARAM_COUNTS = {
     "resnet50": 25_557_032,
     "resnet101": 44_549_160,
     "resnet152": 60_192_808,
+    "resnext50_32x4d": 25_028_904,
+    "wide_resnet50_2": 68_883_240,
 }
 
 
@@ -23,8 +25,7 @@ def test_registry():
         build_model("nope")
 
 
-@pytest.mark.parametrize("arch", ["resnet18", "resnet34", "resnet50",
-                                  "resnet101", "resnet152"])
+@pytest.mark.parametrize("arch", sorted(PARAM_COUNTS))
 def test_param_counts(arch):
     m = build_model(arch)
     assert sum(p.numel() for p in m.parameters()) == PARAM_COUNTS[arch]
