@@ -67,6 +67,14 @@ class _AmpHandle:
         self.model_params: List[torch.nn.Parameter] = []
         self.master_params: List[torch.Tensor] = []
 
+    @torch.no_grad()
+    def resync_masters(self) -> None:
+        """Re-copy model half weights into the fp32 masters.  Needed after
+        any external overwrite of the model weights (e.g. the DDP rank-0
+        broadcast, which runs after ``initialize`` created the masters)."""
+        for mp, hp in zip(self.master_params, self.model_params):
+            mp.data.copy_(hp.data.to(mp.dtype))
+
 
 def _needs_scaling(dtype: torch.dtype) -> bool:
     return dtype == torch.float16  # bf16 has fp32's exponent range

@@ -32,6 +32,10 @@ class NativeDDP(nn.Module):
         # replicas start identical: rank-0 broadcast of params + buffers
         # (torch-DDP construction broadcast)
         broadcast_module_state(module, src=0)
+        amp_handle = getattr(module, "_amp_handle", None)
+        if amp_handle is not None and amp_handle.master_params:
+            # O2 masters were cloned before the broadcast — re-sync them
+            amp_handle.resync_masters()
         self.reducer = BucketedReducer(
             list(module.parameters()), bucket_cap_mb=bucket_cap_mb,
             compression=compression, process_group=process_group)

@@ -143,7 +143,7 @@ def main():
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step()
+        loss = step()
     comm.barrier()
     torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
@@ -180,6 +180,9 @@ def main():
                 "style": args.style,
                 "sec_per_epoch_est": round(1_281_167 / ips, 1),
                 "hip_ext": not args.no_ext,
+                "peak_mem_gb": round(
+                    torch.cuda.max_memory_allocated() / 2**30, 2),
+                "final_loss": round(float(loss.item()), 4),
             },
         }
         print(json.dumps(result), flush=True)
