@@ -50,8 +50,11 @@ ce_fwd_kernel(const T* __restrict__ logits, const long* __restrict__ target,
 template <typename T>
 __global__ void __launch_bounds__(AMD_TPB)
 ce_bwd_kernel(const T* __restrict__ logits, const long* __restrict__ target,
-              const float* __restrict__ lse, T* __restrict__ grad, float gscale,
-              long B, long C) {
+              const float* __restrict__ lse, T* __restrict__ grad,
+              const float* __restrict__ upstream, float inv_B, long B,
+              long C) {
+  // upstream grad is read on-device (no host .item() sync; graph-capturable)
+  const float gscale = upstream[0] * inv_B;
   const long total = B * C;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
@@ -88,10 +91,13 @@ std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits,
 }
 
 at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor target,
-                             at::Tensor lse, double gscale) {
+                             at::Tensor lse, at::Tensor upstream) {
+  // upstream: scalar fp32 tensor on device (the grad flowing into the mean
+  // loss); the kernel folds in the 1/B of the mean reduction itself
   auto lc = logits.contiguous();
   long B = lc.size(0), C = lc.size(1);
   auto grad = at::empty_like(lc);
+  auto up = upstream.to(at::kFloat).contiguous();
   auto stream = at::cuda::getCurrentCUDAStream();
   int grid = amd_grid(B * C);
   AT_DISPATCH_FLOATING_TYPES_AND2(
@@ -101,7 +107,7 @@ at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor target,
         ce_bwd_kernel<devT><<<grid, AMD_TPB, 0, stream>>>(
             (const devT*)lc.const_data_ptr(), target.data_ptr<long>(),
             lse.data_ptr<float>(), (devT*)grad.data_ptr(),
-            (float)gscale, B, C);
+            up.data_ptr<float>(), 1.f / (float)B, B, C);
         CHECK_CUDA_OK();
       });
   return grad;

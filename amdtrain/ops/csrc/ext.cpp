@@ -14,7 +14,7 @@ void multi_tensor_sgd(std::vector<at::Tensor> params,
 // cross_entropy.hip
 std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor target);
 at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor target,
-                             at::Tensor lse, double gscale);
+                             at::Tensor lse, at::Tensor upstream);
 
 // elementwise.hip
 at::Tensor normalize_u8(at::Tensor x, std::vector<double> mean,

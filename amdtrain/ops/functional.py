@@ -42,12 +42,9 @@ class _CrossEntropy(torch.autograd.Function):
     def backward(ctx, grad_out: torch.Tensor):
         logits, target, lse = ctx.saved_tensors
         e = require_ext()
-        gscale = grad_out.item() / logits.size(0) if grad_out.numel() == 1 else None
-        if gscale is not None:
-            grad_logits = e.cross_entropy_bwd(logits, target, lse, gscale)
-        else:  # non-scalar upstream grad: fall back (never hit in training)
-            grad_logits = e.cross_entropy_bwd(logits, target, lse, 1.0 / logits.size(0))
-            grad_logits = grad_logits * grad_out
+        # upstream grad stays on device: no host sync per backward
+        grad_logits = e.cross_entropy_bwd(logits, target, lse,
+                                          grad_out.reshape(1))
         return grad_logits, None
 
 
