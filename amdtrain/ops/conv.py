@@ -6,10 +6,9 @@ pre-transposed weight, wgrad is the TN GEMM with split-M accumulation.
 Strided (downsample) 1x1 convs gather/scatter the even rows around the same
 GEMMs.
 
-Dispatch: ``AmdConv2d`` routes eligible 1x1 convs (GPU + extension + bf16
-path + AMDTRAIN_CONV1X1=custom) to these kernels, everything else to MIOpen
-via the standard conv path.  The default is chosen from measurement — see
-profiles/ for the per-shape comparison.
+Dispatch: ``AmdConv2d`` routes all eligible convs (GPU + extension + bf16
+path) to the hand-written kernels; see the class docstring for the matrix.
+profiles/ holds the per-shape measurements behind the defaults.
 """
 
 from __future__ import annotations
@@ -169,11 +168,15 @@ def _conv1x1_env_default() -> str:
 
 
 class AmdConv2d(nn.Conv2d):
-    """nn.Conv2d whose eligible 1x1 instances run the MFMA GEMM path.
+    """nn.Conv2d dispatching to the hand-written gfx950 kernels.
 
-    State dict / init are identical to nn.Conv2d.  Non-1x1 shapes (3x3
-    spatial convs, the 7x7 stem) currently run through MIOpen; their
-    implicit-GEMM HIP kernels are the next build stage.
+    State dict / init are identical to nn.Conv2d.  On-GPU bf16 paths:
+    1x1 -> MFMA GEMM (gemm.hip), 3x3 pad-1 s1/s2 -> implicit GEMM
+    (conv3x3.hip), anything else without an input gradient (the 7x7
+    stem) -> generic element-gather implicit GEMM (conv_stem.hip).
+    Grouped/dilated convs and fp32 inference fall through to MIOpen.
+    Env overrides AMDTRAIN_CONV1X1 / AMDTRAIN_CONV3X3 / AMDTRAIN_CONVSTEM
+    = "miopen" for A/B measurement.
     """
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

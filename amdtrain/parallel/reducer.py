@@ -189,9 +189,14 @@ class BucketedReducer:
             self._grad_accum = prev
 
     def zero_grad(self) -> None:
-        """Zero all bucket buffers (== zeroing every param.grad view)."""
+        """Zero all bucket buffers (== zeroing every param.grad view) and
+        reset per-iteration state (robust to an aborted backward)."""
         for b in self.buckets:
             b.flat.zero_()
+            b.ready = 0
+            b.work = None
+        self._launch_order.clear()
+        self._cb_queued = False
 
     def grad_buffers(self) -> List[torch.Tensor]:
         return [b.flat for b in self.buckets]
