@@ -11,7 +11,7 @@ from __future__ import annotations
 
 import math
 import random
-from typing import Sequence, Tuple
+from typing import Sequence
 
 import numpy as np
 import torch

@@ -14,7 +14,6 @@ profiles/ holds the per-shape measurements behind the defaults.
 from __future__ import annotations
 
 import os
-from typing import Optional
 
 import torch
 import torch.nn as nn

@@ -11,7 +11,7 @@ launches through cuDNN/cuBLAS/apex/horovod).
 
 from __future__ import annotations
 
-from typing import Iterable, List, Optional, Sequence, Tuple
+from typing import List, Sequence
 
 import torch
 import torch.nn.functional as F

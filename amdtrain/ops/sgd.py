@@ -13,7 +13,7 @@ Semantics match ``torch.optim.SGD`` (momentum buffer ``b = mu*b + g + wd*p``,
 
 from __future__ import annotations
 
-from typing import Iterable, List
+from typing import List
 
 import torch
 from torch.optim import Optimizer

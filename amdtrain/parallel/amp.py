@@ -21,7 +21,7 @@ inf/nan (step skipped), growth x2 every 2000 clean steps.
 from __future__ import annotations
 
 import contextlib
-from typing import List, Optional
+from typing import List
 
 import torch
 

@@ -10,8 +10,6 @@ explicit ``finish()`` call is needed, same usage contract as torch DDP.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.nn as nn
 

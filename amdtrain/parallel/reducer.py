@@ -36,8 +36,6 @@ from typing import Dict, List, Optional, Sequence
 import torch
 import torch.distributed as dist
 
-from ..ops import functional as OF
-
 
 class _Bucket:
     __slots__ = ("params", "flat", "comm_buf", "ready", "work", "index")

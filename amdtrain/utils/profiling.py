@@ -8,7 +8,7 @@ table for a few training steps — the in-framework complement to out-of-band
 from __future__ import annotations
 
 import contextlib
-from typing import Callable, Optional
+from typing import Optional
 
 import torch
 
