@@ -103,11 +103,17 @@ def run_worker(local_rank: int, nprocs: int, args, style: StyleConfig,
         args, world_size=world, rank=rank,
         distributed=(world > 1), distributed_val=style.shard_val)
 
+    # activation input dtype: the half dtype for O2 (model weights are
+    # half), the autocast dtype otherwise
+    data_dtype = None
+    if style.amp_level == "O2" and device.type == "cuda":
+        data_dtype = _DTYPES[args.dtype] or torch.bfloat16
     state = TrainState(
         device=device, world_size=world, rank=rank,
         reduce_metrics=style.reduce_metrics,
         channels_last=channels_last,
         autocast_dtype=autocast_dtype,
+        data_dtype=data_dtype,
         print_freq=args.print_freq,
         max_steps=args.max_steps,
     )
@@ -143,6 +149,6 @@ def _wrap_loader(loader, style: StyleConfig, state: TrainState):
     if style.use_prefetcher and state.device.type == "cuda" \
             and not getattr(loader, "_amdtrain_prefetched", False):
         return CudaPrefetcher(loader, device=state.device,
-                              dtype=state.autocast_dtype or torch.float32,
+                              dtype=state.input_dtype(),
                               channels_last=state.channels_last)
     return loader

@@ -37,8 +37,12 @@ class TrainState:
     reduce_metrics: bool = True      # slurm-style sets False
     channels_last: bool = True
     autocast_dtype: Optional[torch.dtype] = None  # None => fp32
+    data_dtype: Optional[torch.dtype] = None      # activation input dtype
     print_freq: int = 10
     max_steps: int = 0               # 0 = full epoch
+
+    def input_dtype(self) -> torch.dtype:
+        return self.data_dtype or self.autocast_dtype or torch.float32
 
 
 def _autocast(state: TrainState, model):
@@ -59,8 +63,7 @@ def _to_device(batch, state: TrainState):
         images = images.contiguous(memory_format=torch.channels_last)
     if images.dtype == torch.uint8:
         from ..ops import functional as OF
-        images = OF.normalize_u8(
-            images, dtype=state.autocast_dtype or torch.float32)
+        images = OF.normalize_u8(images, dtype=state.input_dtype())
     return images, target
 
 
