@@ -65,13 +65,17 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", 1))
     rank = int(os.environ.get("RANK", 0))
     local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    cpu_mode = os.environ.get("AMDTRAIN_BENCH_CPU") == "1"  # test-only
     if world > 1:
-        comm.init_from_env()
+        comm.init_from_env(backend="gloo" if cpu_mode else None)
 
-    assert torch.cuda.is_available(), "bench.py requires an MI355X"
-    torch.cuda.set_device(local_rank)
-    device = torch.device(f"cuda:{local_rank}")
-    torch.backends.cudnn.benchmark = True
+    if cpu_mode:
+        device = torch.device("cpu")
+    else:
+        assert torch.cuda.is_available(), "bench.py requires an MI355X"
+        torch.cuda.set_device(local_rank)
+        device = torch.device(f"cuda:{local_rank}")
+        torch.backends.cudnn.benchmark = True
 
     B = args.batch_per_gpu
     S = args.image_size
@@ -131,6 +135,10 @@ def main():
         optimizer.step()
         return loss
 
+    def sync():
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+
     for _ in range(args.warmup):
         step()
     if args.profile and rank == 0:
@@ -138,14 +146,14 @@ def main():
         with profile_steps(args.profile):
             for _ in range(3):
                 step()
-        torch.cuda.synchronize()
+        sync()
     comm.barrier()
-    torch.cuda.synchronize()
+    sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         loss = step()
     comm.barrier()
-    torch.cuda.synchronize()
+    sync()
     elapsed = time.perf_counter() - t0
 
     # MAX over ranks
@@ -181,7 +189,8 @@ def main():
                 "sec_per_epoch_est": round(1_281_167 / ips, 1),
                 "hip_ext": not args.no_ext,
                 "peak_mem_gb": round(
-                    torch.cuda.max_memory_allocated() / 2**30, 2),
+                    torch.cuda.max_memory_allocated() / 2**30, 2)
+                if device.type == "cuda" else None,
                 "final_loss": round(float(loss.item()), 4),
             },
         }

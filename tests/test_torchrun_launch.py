@@ -50,3 +50,27 @@ def test_torchrun_horovod_style_entrypoint(tmp_path):
                   ["--compression", "none"], tmp_path, 29615)
     assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
     assert (tmp_path / "checkpoint.pth.tar").exists()
+
+
+def test_torchrun_bench_contract_cpu(tmp_path):
+    """Run bench.py exactly as the scaling driver does (torchrun, 2 procs) —
+    CPU mode via AMDTRAIN_BENCH_CPU=1 — and validate the JSON contract."""
+    import json
+    cmd = [sys.executable, "-m", "torch.distributed.run",
+           "--nnodes=1", "--nproc-per-node=2",
+           "--master-addr", "127.0.0.1", "--master-port", "29617",
+           os.path.join(ROOT, "bench.py"), "--gpus", "2", "--steps", "2",
+           "--warmup", "1", "--batch-per-gpu", "2", "--arch", "resnet18",
+           "--image-size", "32", "--dtype", "fp32"]
+    env = dict(os.environ, AMDTRAIN_BENCH_CPU="1",
+               AMDTRAIN_DISABLE_EXT="1", AMDTRAIN_ALLOW_EAGER="1",
+               PYTHONPATH=ROOT + os.pathsep + os.environ.get("PYTHONPATH", ""))
+    r = subprocess.run(cmd, cwd=str(tmp_path), env=env, capture_output=True,
+                       text=True, timeout=420)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["n_gpus"] == 2 and d["steps"] == 2
+    assert d["unit"] == "images/sec" and d["higher_is_better"] is True
+    assert d["config"]["global_batch"] == 4
+    assert d["scaling"] == "weak" and d["vs_baseline"] is not None
