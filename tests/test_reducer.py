@@ -106,3 +106,14 @@ def test_reducer_accumulation_across_backwards():
     model(x).sum().backward()  # accumulates into the same views
     for p, g in zip(model.parameters(), g1):
         assert torch.allclose(p.grad, 2 * g, atol=1e-5)
+
+
+def test_reducer_four_ranks():
+    """4-rank gloo: averaged gradients with multiple buckets (rank math at
+    higher world sizes, closer to the 8-GPU node)."""
+    res = run_distributed(_reducer_fn, world=4, args=("none",))
+    grads = [_local_grads(100 + r) for r in range(4)]
+    expected = [sum(gs) / 4 for gs in zip(*grads)]
+    for rank in range(4):
+        for got, want in zip(res[rank], expected):
+            assert torch.allclose(got, want, atol=1e-6)
