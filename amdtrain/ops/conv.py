@@ -35,17 +35,21 @@ class _Conv1x1(torch.autograd.Function):
         n, cin, h, w = x.shape
         cout = weight.shape[0]
         xc = x.contiguous(memory_format=torch.channels_last)
-        if stride > 1:
-            xc = xc[:, :, ::stride, ::stride] \
-                .contiguous(memory_format=torch.channels_last)
-        ho, wo = xc.shape[2], xc.shape[3]
         x2d = _rows(xc)
         w2d = weight.reshape(cout, cin)
-        y2d = e.gemm_bt(x2d, w2d, False)
+        if stride > 1:
+            # even-row gather happens inside the A staging (no copy); the
+            # full x2d is saved — it is the same tensor the sibling conv of
+            # the residual block already saves, so no extra memory
+            y2d = e.gemm_bt_strided(x2d, w2d, n, h, w, stride)
+            ho = (h + stride - 1) // stride
+            wo = (w + stride - 1) // stride
+        else:
+            y2d = e.gemm_bt(x2d, w2d, False)
+            ho, wo = h, w
         ctx.save_for_backward(x2d, w2d)
         ctx.meta = (n, cin, h, w, stride, ho, wo, cout)
-        y = y2d.view(n, ho, wo, cout).permute(0, 3, 1, 2)
-        return y
+        return y2d.view(n, ho, wo, cout).permute(0, 3, 1, 2)
 
     @staticmethod
     @torch.amp.custom_bwd(device_type="cuda")
@@ -61,11 +65,11 @@ class _Conv1x1(torch.autograd.Function):
         if stride > 1:
             # fused zero+scatter (one write pass; stride-2 only in ResNet)
             dx = e.scatter_rows_x2(dx2d, n, h, w, ho, wo)
+            dw = e.gemm_tn_strided(gy2d, x2d, n, h, w, stride)
         else:
             dx = dx2d.view(n, ho, wo, cin).permute(0, 3, 1, 2)
-        # wgrad: dW[Cout,Cin] = dY^T x X (fp32 split-M accumulation)
-        dw = e.gemm_tn(gy2d, x2d, 0).reshape(cout, cin, 1, 1)
-        return dx, dw, None
+            dw = e.gemm_tn(gy2d, x2d, 0)
+        return dx, dw.reshape(cout, cin, 1, 1), None
 
 
 def conv1x1_mfma(x: torch.Tensor, weight: torch.Tensor,

@@ -44,6 +44,10 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
 // gemm.hip
 at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out);
 at::Tensor gemm_tn(at::Tensor dY, at::Tensor X, long msplit);
+at::Tensor gemm_bt_strided(at::Tensor A, at::Tensor B, long Nn, long H,
+                           long W, long stride);
+at::Tensor gemm_tn_strided(at::Tensor dY, at::Tensor X, long Nn, long H,
+                           long W, long stride);
 at::Tensor transpose_2d(at::Tensor x);
 
 // conv3x3.hip
@@ -92,6 +96,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("f32_out") = false);
   m.def("gemm_tn", &gemm_tn, py::arg("dY"), py::arg("X"),
         py::arg("msplit") = 0);
+  m.def("gemm_bt_strided", &gemm_bt_strided);
+  m.def("gemm_tn_strided", &gemm_tn_strided);
   m.def("transpose_2d", &transpose_2d);
   m.def("conv3x3_fwd", &conv3x3_fwd);
   m.def("conv3x3_dgrad", &conv3x3_dgrad);

@@ -65,17 +65,46 @@ __device__ __forceinline__ void stage_tile_128x32(
   }
 }
 
+// same, with explicit per-unit source rows (strided 1x1-conv gather)
+__device__ __forceinline__ void stage_tile_rows(
+    const bf16* __restrict__ g, long ld, const long* arow, long k0,
+    bf16* lds) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int rnd = 0; rnd < 2; ++rnd) {
+    int unit = rnd * GEMM_TPB + t;
+    int koff = (unit & 3) * 8;
+    const bf16* src = g + arow[rnd] * ld + k0 + koff;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)(lds + unit * 8), 16,
+        0, 0);
+  }
+}
+
+// compact output row m=(n,ho,wo) -> strided input row (n, ho*s, wo*s)
+struct StrideMap {
+  int H, W, Hout, Wout, stride;  // input spatial dims + output dims
+};
+
+__device__ __forceinline__ long stride_row(long m, const StrideMap& sm) {
+  long t = m;
+  const int wo = (int)(t % sm.Wout); t /= sm.Wout;
+  const int ho = (int)(t % sm.Hout); t /= sm.Hout;
+  return (t * sm.H + (long)ho * sm.stride) * sm.W + (long)wo * sm.stride;
+}
+
 // C/D fragment mapping for mfma_f32_16x16x32_bf16 (guide §3, m89-verified):
 //   col = lane & 15, row = (lane >> 4) * 4 + reg
 // A fragment: lane holds A[row = lane&15][k = (lane>>4)*8 + j]
 // B fragment: lane holds B[k = (lane>>4)*8 + j][col = lane&15]
 //   (from LDS B^T tile [n][k] this is the same contiguous 16B read as A)
 
-template <bool F32OUT>
+template <bool F32OUT, bool STRIDED>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                void* __restrict__ C, long M, long N, long K, int nbm,
-               int nbn) {
+               int nbn, StrideMap sm) {
   __shared__ bf16 As[BM * BK];
   __shared__ bf16 Bs[BN * BK];
 
@@ -98,10 +127,23 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+  long arow[2];
+  if (STRIDED) {  // per-unit gather rows are K-invariant: hoist
+#pragma unroll
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      long m = m0 + (((rnd * GEMM_TPB) + t) >> 2);
+      if (m >= M) m = M - 1;
+      arow[rnd] = stride_row(m, sm);
+    }
+  }
+
   const long ksteps = K / BK;
   for (long kt = 0; kt < ksteps; ++kt) {
     __syncthreads();  // previous compute done before overwriting LDS
-    stage_tile_128x32(A, K, m0, M, kt * BK, As);
+    if (STRIDED)
+      stage_tile_rows(A, K, arow, kt * BK, As);
+    else
+      stage_tile_128x32(A, K, m0, M, kt * BK, As);
     stage_tile_128x32(B, K, n0, N, kt * BK, Bs);
     __syncthreads();  // barrier drains the global_load_lds queue
 
@@ -184,10 +226,36 @@ __device__ __forceinline__ void tn_stage(const bf16* __restrict__ g, long ld,
   }
 }
 
+// tn_stage over strided-gathered rows (row = stride_row(m))
+__device__ __forceinline__ void tn_stage_strided(
+    const bf16* __restrict__ g, long ld, long m0, long M, long col0,
+    const StrideMap& sm, bf16* lds) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int rnd = 0; rnd < 2; ++rnd) {
+    const int unit = rnd * GEMM_TPB + t;
+    const long m = m0 + (unit >> 4);
+    const int c0 = (unit & 15) * 8;
+    bf16 vals[8];
+    if (m < M && col0 + c0 + 8 <= ld) {
+      const long row = stride_row(m, sm);
+      uint4 raw = *(const uint4*)(g + row * ld + col0 + c0);
+      __builtin_memcpy(vals, &raw, 16);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vals[j] = bf16(0.f);
+    }
+    const int mloc = unit >> 4;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) lds[tn_swz(c0 + j, mloc)] = vals[j];
+  }
+}
+
+template <bool STRIDED>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 gemm_tn_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
                float* __restrict__ dW, long M, long N, long K, int nbn,
-               int nbk, int msplit) {
+               int nbk, int msplit, StrideMap sm) {
   __shared__ bf16 Ys[128 * 32];
   __shared__ bf16 Xs[128 * 32];
 
@@ -220,7 +288,10 @@ gemm_tn_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
     const long m0 = mc * 32;
     __syncthreads();
     tn_stage(dY, N, m0, M, n0, Ys);
-    tn_stage(X, K, m0, M, k0, Xs);
+    if (STRIDED)
+      tn_stage_strided(X, K, m0, M, k0, sm, Xs);
+    else
+      tn_stage(X, K, m0, M, k0, Xs);
     __syncthreads();
 
     // A fragment: dY^T[n][m]; B fragment: X^T -> both contiguous b128 reads
@@ -289,14 +360,39 @@ at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out) {
                      Ac.options().dtype(f32_out ? at::kFloat : at::kBFloat16));
   int nbm = (int)((M + BM - 1) / BM), nbn = (int)((N + BN - 1) / BN);
   auto stream = at::cuda::getCurrentCUDAStream();
+  StrideMap sm{0, 0, 0, 0, 1};
   if (f32_out)
-    gemm_bt_kernel<true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+    gemm_bt_kernel<true, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-        C.data_ptr(), M, N, K, nbm, nbn);
+        C.data_ptr(), M, N, K, nbm, nbn, sm);
   else
-    gemm_bt_kernel<false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+    gemm_bt_kernel<false, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-        C.data_ptr(), M, N, K, nbm, nbn);
+        C.data_ptr(), M, N, K, nbm, nbn, sm);
+  CHECK_CUDA_OK();
+  return C;
+}
+
+// strided 1x1-conv forward: y[m=(n,ho,wo), n'] = sum_k x[(n,ho*s,wo*s), k]
+// * w[n', k] — the even-row gather happens inside the A staging (the
+// separate .contiguous() gather copy this replaces cost ~0.5 ms/step)
+at::Tensor gemm_bt_strided(at::Tensor A, at::Tensor B, long Nn, long H,
+                           long W, long stride) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16 &&
+              B.scalar_type() == at::kBFloat16);
+  auto Ac = A.contiguous();
+  auto Bc = B.contiguous();
+  long K = Ac.size(1), N = Bc.size(0);
+  TORCH_CHECK(Bc.size(1) == K && K % BK == 0);
+  long Hout = (H + stride - 1) / stride, Wout = (W + stride - 1) / stride;
+  long M = Nn * Hout * Wout;
+  auto C = at::empty({M, N}, Ac.options());
+  int nbm = (int)((M + BM - 1) / BM), nbn = (int)((N + BN - 1) / BN);
+  StrideMap sm{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
+  auto stream = at::cuda::getCurrentCUDAStream();
+  gemm_bt_kernel<false, true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+      (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
+      C.data_ptr(), M, N, K, nbm, nbn, sm);
   CHECK_CUDA_OK();
   return C;
 }
@@ -316,9 +412,32 @@ at::Tensor gemm_tn(at::Tensor dY, at::Tensor X, long msplit) {
   if (msplit <= 0)
     msplit = std::max<long>(1, std::min<long>((M + 31) / 32, 512 / tiles));
   auto stream = at::cuda::getCurrentCUDAStream();
-  gemm_tn_kernel<<<(int)(tiles * msplit), GEMM_TPB, 0, stream>>>(
+  StrideMap sm{0, 0, 0, 0, 1};
+  gemm_tn_kernel<false><<<(int)(tiles * msplit), GEMM_TPB, 0, stream>>>(
       (const bf16*)Yc.const_data_ptr(), (const bf16*)Xc.const_data_ptr(),
-      dW.data_ptr<float>(), M, N, K, nbn, nbk, (int)msplit);
+      dW.data_ptr<float>(), M, N, K, nbn, nbk, (int)msplit, sm);
+  CHECK_CUDA_OK();
+  return dW;
+}
+
+// wgrad of the strided 1x1 conv: X rows gathered in-kernel
+at::Tensor gemm_tn_strided(at::Tensor dY, at::Tensor X, long Nn, long H,
+                           long W, long stride) {
+  auto Yc = dY.contiguous();
+  auto Xc = X.contiguous();
+  long M = Yc.size(0), N = Yc.size(1), K = Xc.size(1);
+  TORCH_CHECK(N % 8 == 0 && K % 8 == 0);
+  long Hout = (H + stride - 1) / stride, Wout = (W + stride - 1) / stride;
+  TORCH_CHECK(M == Nn * Hout * Wout, "M mismatch for strided wgrad");
+  auto dW = at::zeros({N, K}, Yc.options().dtype(at::kFloat));
+  int nbn = (int)((N + 127) / 128), nbk = (int)((K + 127) / 128);
+  long tiles = (long)nbn * nbk;
+  long msplit = std::max<long>(1, std::min<long>((M + 31) / 32, 512 / tiles));
+  StrideMap sm{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
+  auto stream = at::cuda::getCurrentCUDAStream();
+  gemm_tn_kernel<true><<<(int)(tiles * msplit), GEMM_TPB, 0, stream>>>(
+      (const bf16*)Yc.const_data_ptr(), (const bf16*)Xc.const_data_ptr(),
+      dW.data_ptr<float>(), M, N, K, nbn, nbk, (int)msplit, sm);
   CHECK_CUDA_OK();
   return dW;
 }
