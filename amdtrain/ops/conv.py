@@ -27,7 +27,15 @@ def _rows(x: torch.Tensor) -> torch.Tensor:
     return x.permute(0, 2, 3, 1).reshape(n * h * w, c)
 
 
+def _fuse_stats_enabled() -> bool:
+    return os.environ.get("AMDTRAIN_FUSE_BNSTATS", "1") == "1"
+
+
 class _Conv1x1(torch.autograd.Function):
+    """1x1 conv; forward also emits per-block BN-statistics partials
+    (non-differentiable 2nd output) for the fused conv->BN pipeline when
+    stride == 1 and fusion is enabled."""
+
     @staticmethod
     @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
     def forward(ctx, x: torch.Tensor, weight: torch.Tensor, stride: int):
@@ -37,6 +45,7 @@ class _Conv1x1(torch.autograd.Function):
         xc = x.contiguous(memory_format=torch.channels_last)
         x2d = _rows(xc)
         w2d = weight.reshape(cout, cin)
+        stats = None
         if stride > 1:
             # even-row gather happens inside the A staging (no copy); the
             # full x2d is saved — it is the same tensor the sibling conv of
@@ -44,16 +53,23 @@ class _Conv1x1(torch.autograd.Function):
             y2d = e.gemm_bt_strided(x2d, w2d, n, h, w, stride)
             ho = (h + stride - 1) // stride
             wo = (w + stride - 1) // stride
+        elif _fuse_stats_enabled():
+            y2d, stats = e.gemm_bt_stats(x2d, w2d)
+            ho, wo = h, w
         else:
             y2d = e.gemm_bt(x2d, w2d, False)
             ho, wo = h, w
         ctx.save_for_backward(x2d, w2d)
         ctx.meta = (n, cin, h, w, stride, ho, wo, cout)
-        return y2d.view(n, ho, wo, cout).permute(0, 3, 1, 2)
+        y = y2d.view(n, ho, wo, cout).permute(0, 3, 1, 2)
+        if stats is None:
+            stats = torch.empty(0, device=x.device)
+        ctx.mark_non_differentiable(stats)
+        return y, stats
 
     @staticmethod
     @torch.amp.custom_bwd(device_type="cuda")
-    def backward(ctx, grad_y: torch.Tensor):
+    def backward(ctx, grad_y: torch.Tensor, _grad_stats=None):
         e = require_ext()
         x2d, w2d = ctx.saved_tensors
         n, cin, h, w, stride, ho, wo, cout = ctx.meta
@@ -74,7 +90,10 @@ class _Conv1x1(torch.autograd.Function):
 
 def conv1x1_mfma(x: torch.Tensor, weight: torch.Tensor,
                  stride: int = 1) -> torch.Tensor:
-    return _Conv1x1.apply(x, weight, stride)
+    y, stats = _Conv1x1.apply(x, weight, stride)
+    if stats.numel():
+        y._amdtrain_bn_stats = stats
+    return y
 
 
 class _Conv3x3(torch.autograd.Function):
@@ -91,16 +110,22 @@ class _Conv3x3(torch.autograd.Function):
         # channels_last weight [Cout,Cin,3,3] is [Cout][kh][kw][Cin] in memory
         w2d = weight.contiguous(memory_format=torch.channels_last) \
             .permute(0, 2, 3, 1).reshape(cout, 9 * cin)
-        y2d = e.conv3x3_fwd(x2d, n, h, w, stride, w2d)
+        if _fuse_stats_enabled():
+            y2d, stats = e.conv3x3_fwd_stats(x2d, n, h, w, stride, w2d)
+        else:
+            y2d = e.conv3x3_fwd(x2d, n, h, w, stride, w2d)
+            stats = torch.empty(0, device=x.device)
         ctx.save_for_backward(x2d, w2d)
         ctx.meta = (n, cin, h, w, stride, cout)
         ho = (h + 2 - 3) // stride + 1
         wo = (w + 2 - 3) // stride + 1
-        return y2d.view(n, ho, wo, cout).permute(0, 3, 1, 2)
+        y = y2d.view(n, ho, wo, cout).permute(0, 3, 1, 2)
+        ctx.mark_non_differentiable(stats)
+        return y, stats
 
     @staticmethod
     @torch.amp.custom_bwd(device_type="cuda")
-    def backward(ctx, grad_y: torch.Tensor):
+    def backward(ctx, grad_y: torch.Tensor, _grad_stats=None):
         e = require_ext()
         x2d, w2d = ctx.saved_tensors
         n, cin, h, w, stride, cout = ctx.meta
@@ -117,7 +142,10 @@ class _Conv3x3(torch.autograd.Function):
 
 def conv3x3_mfma(x: torch.Tensor, weight: torch.Tensor,
                  stride: int = 1) -> torch.Tensor:
-    return _Conv3x3.apply(x, weight, stride)
+    y, stats = _Conv3x3.apply(x, weight, stride)
+    if stats.numel():
+        y._amdtrain_bn_stats = stats
+    return y
 
 
 class _ConvGeneric(torch.autograd.Function):

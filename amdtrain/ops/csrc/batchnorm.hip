@@ -402,6 +402,65 @@ std::vector<at::Tensor> batch_norm_fwd_train(
   return {y, mean, invstd};
 }
 
+// forward-train using per-block statistics partials produced by the conv
+// epilogue (conv fused BN-stats: the separate reduce pass over x is skipped)
+std::vector<at::Tensor> batch_norm_fwd_train_from_parts(
+    at::Tensor x, at::Tensor parts, at::Tensor weight, at::Tensor bias,
+    at::Tensor running_mean, at::Tensor running_var, double momentum,
+    double eps, bool relu, std::optional<at::Tensor> addend) {
+  check_nhwc(x);
+  const long C = x.size(1);
+  const long R = x.numel() / C;
+  TORCH_CHECK(parts.size(1) == 2 * C, "parts must be [nparts][2C]");
+  const int nparts = (int)parts.size(0);
+  auto opts = x.options().dtype(at::kFloat);
+  auto mean = at::empty({C}, opts);
+  auto invstd = at::empty({C}, opts);
+  auto scale = at::empty({C}, opts);
+  auto shift = at::empty({C}, opts);
+  auto y = at::empty_like(x);
+  auto stream = at::cuda::getCurrentCUDAStream();
+
+  auto sums2 = at::empty({BN_COLLAPSE, 2 * C}, opts);
+  {
+    int cgrid = BN_COLLAPSE * (int)((2 * C + AMD_TPB - 1) / AMD_TPB);
+    bn_collapse_partials_kernel<<<cgrid, AMD_TPB, 0, stream>>>(
+        parts.data_ptr<float>(), sums2.data_ptr<float>(), nparts,
+        (int)(2 * C));
+    CHECK_CUDA_OK();
+  }
+  int fgrid = (int)((C + AMD_TPB - 1) / AMD_TPB);
+  bn_finalize_train_kernel<<<fgrid, AMD_TPB, 0, stream>>>(
+      sums2.data_ptr<float>(), BN_COLLAPSE, weight.data_ptr<float>(),
+      bias.data_ptr<float>(), running_mean.data_ptr<float>(),
+      running_var.data_ptr<float>(), mean.data_ptr<float>(),
+      invstd.data_ptr<float>(), scale.data_ptr<float>(),
+      shift.data_ptr<float>(), R, (int)C, (float)momentum, (float)eps);
+  CHECK_CUDA_OK();
+  dispatch_vec(x, [&](auto* tp, auto vec) {
+    using devT = std::remove_pointer_t<decltype(tp)>;
+    constexpr int VEC = decltype(vec)::value;
+    long total_vec = R * C / VEC;
+    int agrid = amd_grid(total_vec);
+    size_t smem = 2 * C * sizeof(float);
+    const devT* zp =
+        addend ? (const devT*)addend->const_data_ptr() : nullptr;
+#define APPLY2(RELU_, ADD_)                                                 \
+  bn_apply_kernel<devT, VEC, RELU_, ADD_>                                   \
+      <<<agrid, AMD_TPB, smem, stream>>>(                                   \
+          (const devT*)x.const_data_ptr(), zp, (devT*)y.data_ptr(),         \
+          scale.data_ptr<float>(), shift.data_ptr<float>(),                 \
+          total_vec, (int)C)
+    if (relu && addend) APPLY2(true, true);
+    else if (relu) APPLY2(true, false);
+    else if (addend) APPLY2(false, true);
+    else APPLY2(false, false);
+#undef APPLY2
+    CHECK_CUDA_OK();
+  });
+  return {y, mean, invstd};
+}
+
 at::Tensor batch_norm_fwd_eval(at::Tensor x, at::Tensor weight,
                                at::Tensor bias, at::Tensor running_mean,
                                at::Tensor running_var, double eps, bool relu,

@@ -149,6 +149,56 @@ __device__ __forceinline__ void stage_plain(
   }
 }
 
+// per-block column (sum, sumsq) of the fp32 accumulator tile (see gemm.hip)
+__device__ __forceinline__ void conv_epilogue_stats(
+    float* __restrict__ stats, const float* acc_flat, long m0, long M,
+    long n0, long N, int bm, int wm, int wn, int fr, int fq,
+    bf16* lds_scratch) {
+  float* srow = stats + (long)bm * 2 * N;
+  float s1[4], s2[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    s1[j] = 0.f;
+    s2[j] = 0.f;
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = m0 + wm + i * 16 + fq * 4 + r;
+        float v = (row < M) ? acc_flat[(i * 4 + j) * 4 + r] : 0.f;
+        s1[j] += v;
+        s2[j] += v * v;
+      }
+  }
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+#pragma unroll
+    for (int off = 32; off >= 16; off >>= 1) {
+      s1[j] += __shfl_down(s1[j], off, AMD_WAVE);
+      s2[j] += __shfl_down(s2[j], off, AMD_WAVE);
+    }
+  }
+  float* red = (float*)lds_scratch;
+  __syncthreads();
+  if (fq == 0) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = wn + j * 16 + fr;
+      red[(wm ? 1 : 0) * 256 + col * 2 + 0] = s1[j];
+      red[(wm ? 1 : 0) * 256 + col * 2 + 1] = s2[j];
+    }
+  }
+  __syncthreads();
+  const int t = threadIdx.x;
+  for (int col = t; col < 128; col += GEMM_TPB) {
+    long n = n0 + col;
+    if (n < N) {
+      srow[n] = red[col * 2] + red[256 + col * 2];
+      srow[N + n] = red[col * 2 + 1] + red[256 + col * 2 + 1];
+    }
+  }
+}
+
 // fwd / dgrad main kernel.  DGRAD only changes the gather map; operand roles:
 //   fwd:   A = x rows (AC channels), B = w [NC, 9*AC], C = y [M, NC]
 //   dgrad: A = dy rows (AC = Cout), B = w' [NC = Cin, 9*Cout], C = dx
@@ -156,7 +206,8 @@ template <bool DGRAD>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
                bf16* __restrict__ C, long M, int AC, int NC, ConvGeom g,
-               int nbm, int nbn, const bf16* __restrict__ zero_page) {
+               int nbm, int nbn, const bf16* __restrict__ zero_page,
+               float* __restrict__ stats) {
   __shared__ bf16 As[128 * BK];
   __shared__ bf16 Bs[128 * BK];
 
@@ -221,6 +272,9 @@ conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
         if (row < M && col < NC)
           C[row * NC + col] = __float2bfloat16(acc[i][j][r]);
       }
+  if (stats != nullptr)
+    conv_epilogue_stats(stats, (const float*)acc, m0, M, n0, NC, bm, wm, wn,
+                        fr, fq, As);
 }
 
 // wgrad per tap: dW[cout, tap*Cin + cin] += sum_m dY[m, cout] * Xg[m, cin]
@@ -352,8 +406,10 @@ static at::Tensor zero_page_for(const at::Tensor& like) {
   return zp;
 }
 
-at::Tensor conv3x3_fwd(at::Tensor x2d, long Nn, long H, long W, long stride,
-                       at::Tensor w2d) {
+std::vector<at::Tensor> conv3x3_fwd_stats_impl(at::Tensor x2d, long Nn,
+                                                long H, long W, long stride,
+                                                at::Tensor w2d,
+                                                bool want_stats) {
   // x2d: [Nn*H*W, Cin] bf16 NHWC rows; w2d: [Cout, 9*Cin]
   TORCH_CHECK(x2d.is_cuda() && x2d.scalar_type() == at::kBFloat16);
   long Cin = x2d.size(1), Cout = w2d.size(0);
@@ -365,13 +421,31 @@ at::Tensor conv3x3_fwd(at::Tensor x2d, long Nn, long H, long W, long stride,
   ConvGeom g{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
   int nbm = (int)((M + 127) / 128), nbn = (int)((Cout + 127) / 128);
   auto zp = zero_page_for(x2d);
+  at::Tensor stats;
+  float* stats_ptr = nullptr;
+  if (want_stats) {
+    stats = at::empty({nbm, 2 * Cout}, x2d.options().dtype(at::kFloat));
+    stats_ptr = stats.data_ptr<float>();
+  }
   auto stream = at::cuda::getCurrentCUDAStream();
   conv3x3_kernel<false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
       (const bf16*)x2d.const_data_ptr(), (const bf16*)w2d.const_data_ptr(),
       (bf16*)y.data_ptr(), M, (int)Cin, (int)Cout, g, nbm, nbn,
-      (const bf16*)zp.const_data_ptr());
+      (const bf16*)zp.const_data_ptr(), stats_ptr);
   CHECK_CUDA_OK();
-  return y;
+  if (want_stats) return {y, stats};
+  return {y};
+}
+
+at::Tensor conv3x3_fwd(at::Tensor x2d, long Nn, long H, long W, long stride,
+                       at::Tensor w2d) {
+  return conv3x3_fwd_stats_impl(x2d, Nn, H, W, stride, w2d, false)[0];
+}
+
+std::vector<at::Tensor> conv3x3_fwd_stats(at::Tensor x2d, long Nn, long H,
+                                          long W, long stride,
+                                          at::Tensor w2d) {
+  return conv3x3_fwd_stats_impl(x2d, Nn, H, W, stride, w2d, true);
 }
 
 at::Tensor conv3x3_dgrad(at::Tensor dy2d, long Nn, long H, long W,
@@ -396,7 +470,7 @@ at::Tensor conv3x3_dgrad(at::Tensor dy2d, long Nn, long H, long W,
   conv3x3_kernel<true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
       (const bf16*)dy2d.const_data_ptr(), (const bf16*)wrot.const_data_ptr(),
       (bf16*)dx.data_ptr(), M, (int)Cout, (int)Cin, g, nbm, nbn,
-      (const bf16*)zp.const_data_ptr());
+      (const bf16*)zp.const_data_ptr(), nullptr);
   CHECK_CUDA_OK();
   return dx;
 }

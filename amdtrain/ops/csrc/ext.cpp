@@ -32,6 +32,10 @@ std::vector<at::Tensor> batch_norm_fwd_train(
     at::Tensor x, at::Tensor weight, at::Tensor bias, at::Tensor running_mean,
     at::Tensor running_var, double momentum, double eps, bool relu,
     std::optional<at::Tensor> addend);
+std::vector<at::Tensor> batch_norm_fwd_train_from_parts(
+    at::Tensor x, at::Tensor parts, at::Tensor weight, at::Tensor bias,
+    at::Tensor running_mean, at::Tensor running_var, double momentum,
+    double eps, bool relu, std::optional<at::Tensor> addend);
 at::Tensor batch_norm_fwd_eval(at::Tensor x, at::Tensor weight,
                                at::Tensor bias, at::Tensor running_mean,
                                at::Tensor running_var, double eps, bool relu,
@@ -53,6 +57,10 @@ at::Tensor transpose_2d(at::Tensor x);
 // conv3x3.hip
 at::Tensor conv3x3_fwd(at::Tensor x2d, long Nn, long H, long W, long stride,
                        at::Tensor w2d);
+std::vector<at::Tensor> conv3x3_fwd_stats(at::Tensor x2d, long Nn, long H,
+                                          long W, long stride,
+                                          at::Tensor w2d);
+std::vector<at::Tensor> gemm_bt_stats(at::Tensor A, at::Tensor B);
 at::Tensor conv3x3_dgrad(at::Tensor dy2d, long Nn, long H, long W,
                          long stride, at::Tensor w2d);
 at::Tensor conv3x3_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn, long H,
@@ -87,6 +95,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
         py::arg("eps"), py::arg("relu"),
         py::arg("addend") = std::nullopt);
+  m.def("batch_norm_fwd_train_from_parts", &batch_norm_fwd_train_from_parts,
+        py::arg("x"), py::arg("parts"), py::arg("weight"), py::arg("bias"),
+        py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
+        py::arg("eps"), py::arg("relu"), py::arg("addend") = std::nullopt);
   m.def("batch_norm_fwd_eval", &batch_norm_fwd_eval,
         py::arg("x"), py::arg("weight"), py::arg("bias"),
         py::arg("running_mean"), py::arg("running_var"), py::arg("eps"),
@@ -100,6 +112,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_tn_strided", &gemm_tn_strided);
   m.def("transpose_2d", &transpose_2d);
   m.def("conv3x3_fwd", &conv3x3_fwd);
+  m.def("conv3x3_fwd_stats", &conv3x3_fwd_stats);
+  m.def("gemm_bt_stats", &gemm_bt_stats);
   m.def("conv3x3_dgrad", &conv3x3_dgrad);
   m.def("conv3x3_wgrad", &conv3x3_wgrad);
   m.def("conv_generic_fwd", &conv_generic_fwd);

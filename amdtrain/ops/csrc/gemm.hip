@@ -100,11 +100,68 @@ __device__ __forceinline__ long stride_row(long m, const StrideMap& sm) {
 // B fragment: lane holds B[k = (lane>>4)*8 + j][col = lane&15]
 //   (from LDS B^T tile [n][k] this is the same contiguous 16B read as A)
 
+// per-block column (sum, sumsq) of the fp32 accumulator tile -> one row of
+// stats[bm][2N]: feeds BN forward without re-reading the conv output.
+// acc layout: lane holds col = base+fr, rows (i*16 + fq*4 + r).
+template <int NFRAG>
+__device__ __forceinline__ void epilogue_stats(
+    float* __restrict__ stats, const float* acc_flat, long m0, long M,
+    long n0, long N, int bm, int wm, int wn, int fr, int fq, int nbm,
+    bf16* lds_scratch) {
+  // acc_flat: [4 m-frags][NFRAG n-frags][4 rows] per lane
+  float* srow = stats + (long)bm * 2 * N;
+  float s1[NFRAG], s2[NFRAG];
+#pragma unroll
+  for (int j = 0; j < NFRAG; ++j) {
+    s1[j] = 0.f;
+    s2[j] = 0.f;
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = m0 + wm + i * 16 + fq * 4 + r;
+        float v = (row < M) ? acc_flat[(i * NFRAG + j) * 4 + r] : 0.f;
+        s1[j] += v;
+        s2[j] += v * v;
+      }
+  }
+  // reduce over the 4 fq lanes of each column (lanes 16 apart)
+#pragma unroll
+  for (int j = 0; j < NFRAG; ++j) {
+#pragma unroll
+    for (int off = 32; off >= 16; off >>= 1) {
+      s1[j] += __shfl_down(s1[j], off, AMD_WAVE);
+      s2[j] += __shfl_down(s2[j], off, AMD_WAVE);
+    }
+  }
+  // cross-wave: the two m-waves (wm 0/64) cover the same columns; stage in
+  // LDS (reusing the staging buffer, fp32 [2 wavesM][N tile=128][2])
+  float* red = (float*)lds_scratch;
+  __syncthreads();  // tile LDS no longer needed for staging
+  if (fq == 0) {
+#pragma unroll
+    for (int j = 0; j < NFRAG; ++j) {
+      int col = wn + j * 16 + fr;
+      red[(wm ? 1 : 0) * 256 + col * 2 + 0] = s1[j];
+      red[(wm ? 1 : 0) * 256 + col * 2 + 1] = s2[j];
+    }
+  }
+  __syncthreads();
+  const int t = threadIdx.x;
+  for (int col = t; col < 128; col += GEMM_TPB) {
+    long n = n0 + col;
+    if (n < N) {
+      srow[n] = red[col * 2] + red[256 + col * 2];
+      srow[N + n] = red[col * 2 + 1] + red[256 + col * 2 + 1];
+    }
+  }
+}
+
 template <bool F32OUT, bool STRIDED>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                void* __restrict__ C, long M, long N, long K, int nbm,
-               int nbn, StrideMap sm) {
+               int nbn, StrideMap sm, float* __restrict__ stats) {
   __shared__ bf16 As[BM * BK];
   __shared__ bf16 Bs[BN * BK];
 
@@ -180,6 +237,9 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
       }
     }
   }
+  if (stats != nullptr)
+    epilogue_stats<4>(stats, (const float*)acc, m0, M, n0, N, bm, wm, wn,
+                      fr, fq, nbm, As);
 }
 
 // ---- TN GEMM for wgrad: dW[N,K] += sum_m dY[m,n] * X[m,k] ----------------
@@ -364,13 +424,33 @@ at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out) {
   if (f32_out)
     gemm_bt_kernel<true, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-        C.data_ptr(), M, N, K, nbm, nbn, sm);
+        C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr);
   else
     gemm_bt_kernel<false, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-        C.data_ptr(), M, N, K, nbm, nbn, sm);
+        C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr);
   CHECK_CUDA_OK();
   return C;
+}
+
+// 1x1 conv forward + fused per-block BN statistics partials [nbm][2N]
+std::vector<at::Tensor> gemm_bt_stats(at::Tensor A, at::Tensor B) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16 &&
+              B.scalar_type() == at::kBFloat16);
+  auto Ac = A.contiguous();
+  auto Bc = B.contiguous();
+  long M = Ac.size(0), K = Ac.size(1), N = Bc.size(0);
+  TORCH_CHECK(Bc.size(1) == K && K % BK == 0);
+  auto C = at::empty({M, N}, Ac.options());
+  int nbm = (int)((M + BM - 1) / BM), nbn = (int)((N + BN - 1) / BN);
+  auto stats = at::empty({nbm, 2 * N}, Ac.options().dtype(at::kFloat));
+  StrideMap sm{0, 0, 0, 0, 1};
+  auto stream = at::cuda::getCurrentCUDAStream();
+  gemm_bt_kernel<false, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+      (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
+      C.data_ptr(), M, N, K, nbm, nbn, sm, stats.data_ptr<float>());
+  CHECK_CUDA_OK();
+  return {C, stats};
 }
 
 // strided 1x1-conv forward: y[m=(n,ho,wo), n'] = sum_k x[(n,ho*s,wo*s), k]
@@ -392,7 +472,7 @@ at::Tensor gemm_bt_strided(at::Tensor A, at::Tensor B, long Nn, long H,
   auto stream = at::cuda::getCurrentCUDAStream();
   gemm_bt_kernel<false, true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
       (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-      C.data_ptr(), M, N, K, nbm, nbn, sm);
+      C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr);
   CHECK_CUDA_OK();
   return C;
 }

@@ -23,15 +23,22 @@ __all__ = ["batch_norm", "bn_add_relu", "max_pool_3x3_s2", "global_avg_pool"]
 
 
 class _BNFunction(torch.autograd.Function):
-    """Fused BN(+add)(+ReLU) training-mode forward/backward on NHWC GPU tensors."""
+    """Fused BN(+add)(+ReLU) training-mode forward/backward on NHWC GPU
+    tensors.  When the producing conv attached per-block statistics
+    partials (conv epilogue fusion), the reduce pass over x is skipped."""
 
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var,
-                momentum, eps, relu, addend):
+                momentum, eps, relu, addend, parts):
         e = require_ext()
-        y, mean, invstd = e.batch_norm_fwd_train(
-            x, weight, bias, running_mean, running_var,
-            float(momentum), float(eps), bool(relu), addend)
+        if parts is not None:
+            y, mean, invstd = e.batch_norm_fwd_train_from_parts(
+                x, parts, weight, bias, running_mean, running_var,
+                float(momentum), float(eps), bool(relu), addend)
+        else:
+            y, mean, invstd = e.batch_norm_fwd_train(
+                x, weight, bias, running_mean, running_var,
+                float(momentum), float(eps), bool(relu), addend)
         ctx.save_for_backward(x, y, weight, mean, invstd)
         ctx.relu = bool(relu)
         ctx.has_addend = addend is not None
@@ -46,7 +53,7 @@ class _BNFunction(torch.autograd.Function):
             y, weight, mean, invstd, ctx.relu, ctx.has_addend)
         grad_addend = ghat if ctx.has_addend else None
         return (grad_x, grad_w, grad_b, None, None, None, None, None,
-                grad_addend)
+                grad_addend, None)
 
 
 def _bn_torch(x: torch.Tensor, bn, relu: bool,
@@ -71,11 +78,13 @@ def batch_norm(x: torch.Tensor, bn, relu: bool = False,
     if bn.training and bn.track_running_stats and bn.num_batches_tracked is not None:
         bn.num_batches_tracked.add_(1)
     if use_ext_for(x):
+        parts = getattr(x, "_amdtrain_bn_stats", None) if bn.training else None
         xc = x.contiguous(memory_format=torch.channels_last)
         ac = None if addend is None else addend.contiguous(memory_format=torch.channels_last)
         if bn.training:
             return _BNFunction.apply(xc, bn.weight, bn.bias, bn.running_mean,
-                                     bn.running_var, bn.momentum, bn.eps, relu, ac)
+                                     bn.running_var, bn.momentum, bn.eps,
+                                     relu, ac, parts)
         return require_ext().batch_norm_fwd_eval(
             xc, bn.weight, bn.bias, bn.running_mean, bn.running_var,
             float(bn.eps), bool(relu), ac)

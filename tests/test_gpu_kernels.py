@@ -313,3 +313,48 @@ def test_train_step_resnet50_bf16():
     torch.cuda.synchronize()
     assert torch.isfinite(loss).item()
     assert all(p.grad is not None for p in m.parameters())
+
+
+@pytest.mark.parametrize("path", ["conv1x1", "conv3x3"])
+def test_conv_bn_fused_stats_parity(path):
+    """conv-epilogue BN statistics == BN's own reduce pass (running stats,
+    mean, output all agree between the fused and unfused paths)."""
+    import os
+    from amdtrain.models.resnet import FusedBatchNorm2d
+    from amdtrain.ops.conv import conv1x1_mfma, conv3x3_mfma
+    _require_ext()
+    torch.manual_seed(0)
+    cin, cout, hw = 64, 128, 28
+    x = torch.randn(4, cin, hw, hw, device=DEV) \
+        .contiguous(memory_format=torch.channels_last)
+    w = (torch.randn(cout, cin, 1, 1, device=DEV) * cin ** -0.5) \
+        if path == "conv1x1" else \
+        (torch.randn(cout, cin, 3, 3, device=DEV) * (9 * cin) ** -0.5)
+    fn = conv1x1_mfma if path == "conv1x1" else conv3x3_mfma
+
+    bn_a = FusedBatchNorm2d(cout).to(DEV)
+    bn_b = FusedBatchNorm2d(cout).to(DEV)
+    bn_b.load_state_dict(bn_a.state_dict())
+
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        y_fused_conv = fn(x, w, 1)
+    assert getattr(y_fused_conv, "_amdtrain_bn_stats", None) is not None
+    y_fused = bn_a.forward_relu(y_fused_conv)
+
+    os.environ["AMDTRAIN_FUSE_BNSTATS"] = "0"
+    try:
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            y_plain_conv = fn(x, w, 1)
+        assert getattr(y_plain_conv, "_amdtrain_bn_stats", None) is None
+        y_plain = bn_b.forward_relu(y_plain_conv)
+    finally:
+        os.environ.pop("AMDTRAIN_FUSE_BNSTATS")
+
+    # stats from the fp32 accumulator vs from the rounded bf16 output:
+    # small tolerance, but running stats and outputs must agree closely
+    assert torch.allclose(bn_a.running_mean, bn_b.running_mean, atol=1e-2,
+                          rtol=1e-2)
+    assert torch.allclose(bn_a.running_var, bn_b.running_var, atol=1e-2,
+                          rtol=1e-2)
+    assert torch.allclose(y_fused.float(), y_plain.float(), atol=0.05,
+                          rtol=0.05), (y_fused - y_plain).abs().max()
