@@ -330,8 +330,10 @@ at::Tensor conv_generic_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn,
   auto dW2 = at::zeros({Cout, g.Kpad}, dy2d.options().dtype(at::kFloat));
   int nbn = (int)((Cout + 127) / 128), nbk = (int)((g.Kpad + 127) / 128);
   long tiles = (long)nbn * nbk;
-  int msplit = (int)std::max<long>(
-      1, std::min<long>((M + 31) / 32, 512 / tiles));
+  static const bool det = std::getenv("AMDTRAIN_DETERMINISTIC") != nullptr;
+  int msplit = det ? 1
+                   : (int)std::max<long>(
+                         1, std::min<long>((M + 31) / 32, 512 / tiles));
   auto stream = at::cuda::getCurrentCUDAStream();
   conv_generic_wgrad_kernel<<<(int)(tiles * msplit), GEMM_TPB, 0, stream>>>(
       (const bf16*)dy2d.const_data_ptr(), (const bf16*)x2d.const_data_ptr(),

@@ -489,7 +489,10 @@ at::Tensor gemm_tn(at::Tensor dY, at::Tensor X, long msplit) {
   auto dW = at::zeros({N, K}, Yc.options().dtype(at::kFloat));
   int nbn = (int)((N + 127) / 128), nbk = (int)((K + 127) / 128);
   long tiles = (long)nbn * nbk;
-  if (msplit <= 0)
+  static const bool det = std::getenv("AMDTRAIN_DETERMINISTIC") != nullptr;
+  if (det)
+    msplit = 1;  // single-accumulator: bitwise-deterministic wgrad
+  else if (msplit <= 0)
     msplit = std::max<long>(1, std::min<long>((M + 31) / 32, 512 / tiles));
   auto stream = at::cuda::getCurrentCUDAStream();
   StrideMap sm{0, 0, 0, 0, 1};
@@ -512,7 +515,9 @@ at::Tensor gemm_tn_strided(at::Tensor dY, at::Tensor X, long Nn, long H,
   auto dW = at::zeros({N, K}, Yc.options().dtype(at::kFloat));
   int nbn = (int)((N + 127) / 128), nbk = (int)((K + 127) / 128);
   long tiles = (long)nbn * nbk;
-  long msplit = std::max<long>(1, std::min<long>((M + 31) / 32, 512 / tiles));
+  static const bool det2 = std::getenv("AMDTRAIN_DETERMINISTIC") != nullptr;
+  long msplit = det2 ? 1 : std::max<long>(
+      1, std::min<long>((M + 31) / 32, 512 / tiles));
   StrideMap sm{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
   auto stream = at::cuda::getCurrentCUDAStream();
   gemm_tn_kernel<true><<<(int)(tiles * msplit), GEMM_TPB, 0, stream>>>(

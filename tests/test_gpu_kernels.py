@@ -358,3 +358,22 @@ def test_conv_bn_fused_stats_parity(path):
                           rtol=1e-2)
     assert torch.allclose(y_fused.float(), y_plain.float(), atol=0.05,
                           rtol=0.05), (y_fused - y_plain).abs().max()
+
+
+def test_deterministic_wgrad(monkeypatch):
+    """AMDTRAIN_DETERMINISTIC=1 (set by --seed) makes conv wgrads bitwise
+    reproducible (single-accumulator split-M)."""
+    monkeypatch.setenv("AMDTRAIN_DETERMINISTIC", "1")
+    from amdtrain import _C
+    _require_ext()
+    torch.manual_seed(0)
+    dY = torch.randn(4096, 64, device=DEV).bfloat16()
+    X = torch.randn(4096, 128, device=DEV).bfloat16()
+    a = _C.gemm_tn(dY, X, 0)
+    b = _C.gemm_tn(dY, X, 0)
+    assert torch.equal(a, b)
+    x2d = torch.randn(2 * 14 * 14, 64, device=DEV).bfloat16()
+    gy2d = torch.randn(2 * 14 * 14, 64, device=DEV).bfloat16()
+    a = _C.conv3x3_wgrad(gy2d, x2d, 2, 14, 14, 1)
+    b = _C.conv3x3_wgrad(gy2d, x2d, 2, 14, 14, 1)
+    assert torch.equal(a, b)
