@@ -485,7 +485,7 @@ at::Tensor conv3x3_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn, long H,
   ConvGeom g{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
   int nbn = (int)((Cout + 127) / 128), nbk = (int)((Cin + 127) / 128);
   long tiles = (long)nbn * nbk;
-  static const bool det = std::getenv("AMDTRAIN_DETERMINISTIC") != nullptr;
+  const bool det = std::getenv("AMDTRAIN_DETERMINISTIC") != nullptr;
   int msplit = det ? 1
                    : (int)std::max<long>(
                          1, std::min<long>((M + 31) / 32, 512 / tiles));

@@ -330,7 +330,7 @@ at::Tensor conv_generic_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn,
   auto dW2 = at::zeros({Cout, g.Kpad}, dy2d.options().dtype(at::kFloat));
   int nbn = (int)((Cout + 127) / 128), nbk = (int)((g.Kpad + 127) / 128);
   long tiles = (long)nbn * nbk;
-  static const bool det = std::getenv("AMDTRAIN_DETERMINISTIC") != nullptr;
+  const bool det = std::getenv("AMDTRAIN_DETERMINISTIC") != nullptr;
   int msplit = det ? 1
                    : (int)std::max<long>(
                          1, std::min<long>((M + 31) / 32, 512 / tiles));

@@ -489,7 +489,7 @@ at::Tensor gemm_tn(at::Tensor dY, at::Tensor X, long msplit) {
   auto dW = at::zeros({N, K}, Yc.options().dtype(at::kFloat));
   int nbn = (int)((N + 127) / 128), nbk = (int)((K + 127) / 128);
   long tiles = (long)nbn * nbk;
-  static const bool det = std::getenv("AMDTRAIN_DETERMINISTIC") != nullptr;
+  const bool det = std::getenv("AMDTRAIN_DETERMINISTIC") != nullptr;
   if (det)
     msplit = 1;  // single-accumulator: bitwise-deterministic wgrad
   else if (msplit <= 0)
@@ -515,7 +515,7 @@ at::Tensor gemm_tn_strided(at::Tensor dY, at::Tensor X, long Nn, long H,
   auto dW = at::zeros({N, K}, Yc.options().dtype(at::kFloat));
   int nbn = (int)((N + 127) / 128), nbk = (int)((K + 127) / 128);
   long tiles = (long)nbn * nbk;
-  static const bool det2 = std::getenv("AMDTRAIN_DETERMINISTIC") != nullptr;
+  const bool det2 = std::getenv("AMDTRAIN_DETERMINISTIC") != nullptr;
   long msplit = det2 ? 1 : std::max<long>(
       1, std::min<long>((M + 31) / 32, 512 / tiles));
   StrideMap sm{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
