@@ -33,7 +33,13 @@ __device__ __forceinline__ int xcd_swz8(int bid, int nwg) {
 }
 
 // stage a [256 rows][64 cols] bf16 tile: 32 KiB = 2048 x 16B units,
-// 4 units per thread; rows are 128 B (8 units per row)
+// 4 units per thread; rows are 128 B (8 units per row).
+//
+// LDS is XOR-swizzled (slot ^= row & 7): the fragment read pattern (16
+// lanes reading 16 consecutive rows at one 16B column) is a 16-way bank
+// conflict on the linear layout.  global_load_lds writes linearly, so the
+// swizzle is applied by PRE-SWIZZLING the per-lane global source and
+// applying the same involution on the ds_read side (guide rule #21).
 __device__ __forceinline__ void stage_256x64(
     const bf16* __restrict__ g, long ld, long row0, long rows, long k0,
     bf16* lds) {
@@ -43,13 +49,18 @@ __device__ __forceinline__ void stage_256x64(
     int unit = rnd * TPB8 + t;           // 0..2047
     long row = row0 + (unit >> 3);       // 8 x 16B units per 128B row
     if (row >= rows) row = rows - 1;
-    int koff = (unit & 7) * 8;           // 8 bf16 per unit
-    const bf16* src = g + row * ld + k0 + koff;
+    int slot = (unit & 7) ^ ((unit >> 3) & 7);  // inverse-swizzled source
+    const bf16* src = g + row * ld + k0 + slot * 8;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)src,
         (__attribute__((address_space(3))) unsigned int*)(lds + unit * 8), 16,
         0, 0);
   }
+}
+
+// swizzled element offset of logical (row, 16B-slot) in a [*][64] tile
+__device__ __forceinline__ int swz8(int row, int slot) {
+  return row * 64 + ((slot ^ (row & 7)) << 3);
 }
 
 __global__ void __launch_bounds__(TPB8, 1)
@@ -89,12 +100,10 @@ gemm_bt_8p_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
       bf16x8 b[4];
 #pragma unroll
       for (int j = 0; j < 4; ++j)
-        b[j] = *(const bf16x8*)&Bs[(wc + j * 16 + fr) * BK8 + kk * 32 +
-                                   fq * 8];
+        b[j] = *(const bf16x8*)&Bs[swz8(wc + j * 16 + fr, kk * 4 + fq)];
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
-        bf16x8 a = *(const bf16x8*)&As[(wr + i * 16 + fr) * BK8 + kk * 32 +
-                                       fq * 8];
+        bf16x8 a = *(const bf16x8*)&As[swz8(wr + i * 16 + fr, kk * 4 + fq)];
 #pragma unroll
         for (int j = 0; j < 4; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -116,7 +125,136 @@ gemm_bt_8p_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
       }
 }
 
+// stage 3: full double-buffer + per-quadrant load/compute interleave.
+// Staging for K-tile kt+1 is issued in 4 half-tile slices between the MFMA
+// quadrants of kt; targets the OTHER buffer, so no LDS hazard and no
+// barrier between phases — one vmcnt drain + barrier per K-tile, covered
+// by 64 MFMA per wave.
+__device__ __forceinline__ void stage_half(
+    const bf16* __restrict__ g, long ld, long row0, long rows, long k0,
+    int half, bf16* lds) {
+  // half h stages rows [h*128, h*128+128) : 1024 units, 2 per thread
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int rnd = 0; rnd < 2; ++rnd) {
+    int unit = half * 1024 + rnd * TPB8 + t;
+    long row = row0 + (unit >> 3);
+    if (row >= rows) row = rows - 1;
+    int slot = (unit & 7) ^ ((unit >> 3) & 7);
+    const bf16* src = g + row * ld + k0 + slot * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)(lds + unit * 8), 16,
+        0, 0);
+  }
+}
+
+__global__ void __launch_bounds__(TPB8, 1)
+gemm_bt_8p3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
+                   bf16* __restrict__ C, long M, long N, long K, int nbm,
+                   int nbn) {
+  __shared__ bf16 As[2][BM8 * BK8];
+  __shared__ bf16 Bs[2][BN8 * BK8];
+
+  const int bid = xcd_swz8(blockIdx.x, nbm * nbn);
+  const int bm = bid / nbn, bn = bid % nbn;
+  const long m0 = (long)bm * BM8, n0 = (long)bn * BN8;
+
+  const int t = threadIdx.x;
+  const int wave = t / AMD_WAVE;
+  const int lane = t % AMD_WAVE;
+  const int wr = (wave >> 2) * 128;
+  const int wc = (wave & 3) * 64;
+  const int fr = lane & 15;
+  const int fq = lane >> 4;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const long ksteps = K / BK8;
+  // prologue: stage K-tile 0 fully
+  stage_256x64(A, K, m0, M, 0, As[0]);
+  stage_256x64(B, K, n0, N, 0, Bs[0]);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (long kt = 0; kt < ksteps; ++kt) {
+    const int cur = (int)(kt & 1), nxt = cur ^ 1;
+    const bool pre = kt + 1 < ksteps;
+    const long k1 = (kt + 1) * BK8;
+
+    // B fragments for the whole K-tile (reused by all quadrants)
+    bf16x8 b[4][2];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        b[j][kk] =
+            *(const bf16x8*)&Bs[cur][swz8(wc + j * 16 + fr, kk * 4 + fq)];
+
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      if (pre) {  // one half-tile of kt+1 between quadrants
+        if (p < 2)
+          stage_half(A, K, m0, M, k1, p, As[nxt]);
+        else
+          stage_half(B, K, n0, N, k1, p - 2, Bs[nxt]);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int iq = 0; iq < 2; ++iq) {
+        const int i = p * 2 + iq;
+        bf16x8 a0 = *(const bf16x8*)&As[cur][swz8(wr + i * 16 + fr, fq)];
+        bf16x8 a1 = *(const bf16x8*)&As[cur][swz8(wr + i * 16 + fr, 4 + fq)];
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a0, b[j][0], acc[i][j], 0, 0, 0);
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a1, b[j][1], acc[i][j], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = m0 + wr + i * 16 + fq * 4 + r;
+        long col = n0 + wc + j * 16 + fr;
+        if (row < M && col < N)
+          C[row * N + col] = __float2bfloat16(acc[i][j][r]);
+      }
+}
+
 }  // namespace
+
+at::Tensor gemm_bt_8p3(at::Tensor A, at::Tensor B) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16 &&
+              B.scalar_type() == at::kBFloat16);
+  auto Ac = A.contiguous();
+  auto Bc = B.contiguous();
+  long M = Ac.size(0), K = Ac.size(1), N = Bc.size(0);
+  TORCH_CHECK(Bc.size(1) == K && K % BK8 == 0);
+  auto C = at::empty({M, N}, Ac.options());
+  int nbm = (int)((M + BM8 - 1) / BM8), nbn = (int)((N + BN8 - 1) / BN8);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  gemm_bt_8p3_kernel<<<nbm * nbn, TPB8, 0, stream>>>(
+      (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
+      (bf16*)C.data_ptr(), M, N, K, nbm, nbn);
+  CHECK_CUDA_OK();
+  return C;
+}
 
 at::Tensor gemm_bt_8p(at::Tensor A, at::Tensor B) {
   TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16 &&

@@ -126,3 +126,18 @@ def test_resnet50_custom_conv1x1_step(monkeypatch):
     opt.step()
     torch.cuda.synchronize()
     assert torch.isfinite(loss).item()
+
+
+@pytest.mark.parametrize("fn_name", ["gemm_bt_8p", "gemm_bt_8p3"])
+def test_gemm_8p_variants_match_matmul(fn_name):
+    """Experimental 256x256-tile kernels (square-GEMM ladder: 761 -> 885 ->
+    954 TF @4096^3) stay correct."""
+    e = _ext()
+    torch.manual_seed(7)
+    for M, N, K in [(512, 512, 128), (1000, 256, 256), (2048, 512, 192)]:
+        A = torch.randn(M, K, device=DEV).bfloat16()
+        B = torch.randn(N, K, device=DEV).bfloat16()
+        C = getattr(e, fn_name)(A, B).float()
+        ref = A.float() @ B.float().t()
+        assert torch.allclose(C, ref, atol=0.5, rtol=0.05), \
+            (C - ref).abs().max()

@@ -26,7 +26,8 @@ def run(M, N, K, check=True):
     if check:
         ref = (A[:512].float() @ B.float().t())
         for name, fn in [("bt", lambda: _C.gemm_bt(A, B, False)),
-                         ("8p", lambda: _C.gemm_bt_8p(A, B))]:
+                         ("8p", lambda: _C.gemm_bt_8p(A, B)),
+                         ("8p3", lambda: _C.gemm_bt_8p3(A, B))]:
             out = fn()[:512].float()
             err = (out - ref).abs().max().item()
             scale = ref.abs().mean().item() + 1e-6
@@ -37,8 +38,10 @@ def run(M, N, K, check=True):
     flops = 2.0 * M * N * K
     t_bt = time_fn(lambda: _C.gemm_bt(A, B, False))
     t_8p = time_fn(lambda: _C.gemm_bt_8p(A, B))
+    t_83 = time_fn(lambda: _C.gemm_bt_8p3(A, B))
     print(f"M={M} N={N} K={K}: bt {t_bt*1e6:8.1f} us ({flops/t_bt/1e12:7.1f} TF)"
-          f" | 8p {t_8p*1e6:8.1f} us ({flops/t_8p/1e12:7.1f} TF)", flush=True)
+          f" | 8p {t_8p*1e6:8.1f} us ({flops/t_8p/1e12:7.1f} TF)"
+          f" | 8p3 {t_83*1e6:8.1f} us ({flops/t_83/1e12:7.1f} TF)", flush=True)
 
 
 if __name__ == "__main__":
