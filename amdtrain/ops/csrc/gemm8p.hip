@@ -149,6 +149,7 @@ __device__ __forceinline__ void stage_half(
   }
 }
 
+template <int ISSUE>  // 0: stage spread across quadrants; 1: all up-front
 __global__ void __launch_bounds__(TPB8, 1)
 gemm_bt_8p3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                    bf16* __restrict__ C, long M, long N, long K, int nbm,
@@ -186,6 +187,11 @@ gemm_bt_8p3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
     const bool pre = kt + 1 < ksteps;
     const long k1 = (kt + 1) * BK8;
 
+    if (ISSUE == 1 && pre) {
+      stage_256x64(A, K, m0, M, k1, As[nxt]);
+      stage_256x64(B, K, n0, N, k1, Bs[nxt]);
+    }
+
     // B fragments for the whole K-tile (reused by all quadrants)
     bf16x8 b[4][2];
 #pragma unroll
@@ -197,7 +203,7 @@ gemm_bt_8p3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
 
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
-      if (pre) {  // one half-tile of kt+1 between quadrants
+      if (ISSUE == 0 && pre) {  // one half-tile of kt+1 between quadrants
         if (p < 2)
           stage_half(A, K, m0, M, k1, p, As[nxt]);
         else
@@ -249,9 +255,16 @@ at::Tensor gemm_bt_8p3(at::Tensor A, at::Tensor B) {
   auto C = at::empty({M, N}, Ac.options());
   int nbm = (int)((M + BM8 - 1) / BM8), nbn = (int)((N + BN8 - 1) / BN8);
   auto stream = at::cuda::getCurrentCUDAStream();
-  gemm_bt_8p3_kernel<<<nbm * nbn, TPB8, 0, stream>>>(
-      (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-      (bf16*)C.data_ptr(), M, N, K, nbm, nbn);
+  // up-front issue measured +3.7% over per-quadrant spread (976 TF @4096^3)
+  const char* iv = std::getenv("AMDTRAIN_8P_ISSUE");
+  if (!(iv && iv[0] == '0'))
+    gemm_bt_8p3_kernel<1><<<nbm * nbn, TPB8, 0, stream>>>(
+        (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
+        (bf16*)C.data_ptr(), M, N, K, nbm, nbn);
+  else
+    gemm_bt_8p3_kernel<0><<<nbm * nbn, TPB8, 0, stream>>>(
+        (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
+        (bf16*)C.data_ptr(), M, N, K, nbm, nbn);
   CHECK_CUDA_OK();
   return C;
 }
