@@ -65,6 +65,8 @@ std::vector<at::Tensor> gemm_bt_stats(at::Tensor A, at::Tensor B);
 // gemm8p.hip
 at::Tensor gemm_bt_8p(at::Tensor A, at::Tensor B);
 at::Tensor gemm_bt_8p3(at::Tensor A, at::Tensor B);
+at::Tensor conv3x3_8p(at::Tensor A2d, long Nn, long H, long W, long stride,
+                      at::Tensor w2d, bool dgrad);
 at::Tensor conv3x3_dgrad(at::Tensor dy2d, long Nn, long H, long W,
                          long stride, at::Tensor w2d);
 at::Tensor conv3x3_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn, long H,
@@ -120,6 +122,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bt_stats", &gemm_bt_stats);
   m.def("gemm_bt_8p", &gemm_bt_8p);
   m.def("gemm_bt_8p3", &gemm_bt_8p3);
+  m.def("conv3x3_8p", &conv3x3_8p);
   m.def("conv3x3_dgrad", &conv3x3_dgrad);
   m.def("conv3x3_wgrad", &conv3x3_wgrad);
   m.def("conv_generic_fwd", &conv_generic_fwd);

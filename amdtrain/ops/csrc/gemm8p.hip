@@ -243,6 +243,177 @@ gemm_bt_8p3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
       }
 }
 
+// ---- conv3x3 on the 8p3 structure (long-K layer3/4 shapes) ---------------
+// Same 256x256 double-buffered schedule; the A staging gathers rows through
+// the 3x3 tap map (pad 1, stride 1/2).  K-tile = 64 channels within ONE tap
+// (Cin % 64 == 0 for all eligible shapes), so the gather row per unit is
+// tap-dependent but fixed across the k-slots of a tile.
+struct Geom8 {
+  int H, W, Hout, Wout, stride;
+};
+
+struct Coord8 {
+  long n_off;
+  int hb, wb;
+};
+
+template <bool DGRAD>
+__device__ __forceinline__ Coord8 dec8(long m, const Geom8& g) {
+  Coord8 u;
+  long t = m;
+  if (DGRAD) {
+    const int w = (int)(t % g.W); t /= g.W;
+    const int h = (int)(t % g.H); t /= g.H;
+    u.n_off = t * (long)g.Hout * g.Wout;
+    u.hb = h + 1;
+    u.wb = w + 1;
+  } else {
+    const int wo = (int)(t % g.Wout); t /= g.Wout;
+    const int ho = (int)(t % g.Hout); t /= g.Hout;
+    u.n_off = t * (long)g.H * g.W;
+    u.hb = ho * g.stride - 1;
+    u.wb = wo * g.stride - 1;
+  }
+  return u;
+}
+
+template <bool DGRAD>
+__device__ __forceinline__ long row8(const Coord8& u, int kh, int kw,
+                                     const Geom8& g) {
+  if (DGRAD) {
+    int ho2 = u.hb - kh, wo2 = u.wb - kw;
+    if (g.stride == 2) {
+      if ((ho2 | wo2) & 1) return -1;
+      ho2 >>= 1;
+      wo2 >>= 1;
+    }
+    if (ho2 < 0 || ho2 >= g.Hout || wo2 < 0 || wo2 >= g.Wout) return -1;
+    return u.n_off + (long)ho2 * g.Wout + wo2;
+  }
+  const int h = u.hb + kh, w = u.wb + kw;
+  if (h < 0 || h >= g.H || w < 0 || w >= g.W) return -1;
+  return u.n_off + (long)h * g.W + w;
+}
+
+// gather-stage a [256 m][64 ch] tile for tap (kh,kw), channel block c0
+template <bool DGRAD>
+__device__ __forceinline__ void stage_conv_256x64(
+    const bf16* __restrict__ g, long ld, const Coord8* uc, int kh, int kw,
+    const Geom8& geo, long c0, const bf16* __restrict__ zp, bf16* lds) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int rnd = 0; rnd < 4; ++rnd) {
+    int unit = rnd * TPB8 + t;
+    long row = row8<DGRAD>(uc[rnd], kh, kw, geo);
+    int slot = (unit & 7) ^ ((unit >> 3) & 7);
+    const bf16* src = row < 0 ? zp : g + row * ld + c0 + slot * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)(lds + unit * 8), 16,
+        0, 0);
+  }
+}
+
+template <bool DGRAD>
+__global__ void __launch_bounds__(TPB8, 1)
+conv3x3_8p_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
+                  bf16* __restrict__ C, long M, int AC, int NC, Geom8 geo,
+                  int nbm, int nbn, const bf16* __restrict__ zp) {
+  __shared__ bf16 As[2][BM8 * BK8];
+  __shared__ bf16 Bs[2][BN8 * BK8];
+
+  const int bid = xcd_swz8(blockIdx.x, nbm * nbn);
+  const int bm = bid / nbn, bn = bid % nbn;
+  const long m0 = (long)bm * BM8, n0 = (long)bn * BN8;
+
+  const int t = threadIdx.x;
+  const int wave = t / AMD_WAVE;
+  const int lane = t % AMD_WAVE;
+  const int wr = (wave >> 2) * 128;
+  const int wc = (wave & 3) * 64;
+  const int fr = lane & 15;
+  const int fq = lane >> 4;
+
+  // hoisted per-unit coords (tap-invariant)
+  Coord8 uc[4];
+#pragma unroll
+  for (int rnd = 0; rnd < 4; ++rnd) {
+    long m = m0 + (((rnd * TPB8) + t) >> 3);
+    if (m >= M) m = M - 1;
+    uc[rnd] = dec8<DGRAD>(m, geo);
+  }
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int cpt = AC / BK8;       // channel K-tiles per tap
+  const long ksteps = 9L * cpt;   // total K-tiles
+
+  // prologue: stage K-tile 0 (tap 0, c0 = 0)
+  stage_conv_256x64<DGRAD>(A, AC, uc, 0, 0, geo, 0, zp, As[0]);
+  stage_256x64(Bw, (long)9 * AC, n0, NC, 0, Bs[0]);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (long kt = 0; kt < ksteps; ++kt) {
+    const int cur = (int)(kt & 1), nxt = cur ^ 1;
+    const bool pre = kt + 1 < ksteps;
+    if (pre) {
+      const long kn = kt + 1;
+      const int tap = (int)(kn / cpt);
+      const long c0 = (kn % cpt) * BK8;
+      stage_conv_256x64<DGRAD>(A, AC, uc, tap / 3, tap % 3, geo, c0, zp,
+                               As[nxt]);
+      stage_256x64(Bw, (long)9 * AC, n0, NC, (long)tap * AC + c0, Bs[nxt]);
+    }
+
+    bf16x8 b[4][2];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        b[j][kk] =
+            *(const bf16x8*)&Bs[cur][swz8(wc + j * 16 + fr, kk * 4 + fq)];
+
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int iq = 0; iq < 2; ++iq) {
+        const int i = p * 2 + iq;
+        bf16x8 a0 = *(const bf16x8*)&As[cur][swz8(wr + i * 16 + fr, fq)];
+        bf16x8 a1 = *(const bf16x8*)&As[cur][swz8(wr + i * 16 + fr, 4 + fq)];
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a0, b[j][0], acc[i][j], 0, 0, 0);
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a1, b[j][1], acc[i][j], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = m0 + wr + i * 16 + fq * 4 + r;
+        long col = n0 + wc + j * 16 + fr;
+        if (row < M && col < NC)
+          C[row * NC + col] = __float2bfloat16(acc[i][j][r]);
+      }
+}
+
 }  // namespace
 
 at::Tensor gemm_bt_8p3(at::Tensor A, at::Tensor B) {
@@ -283,6 +454,37 @@ at::Tensor gemm_bt_8p(at::Tensor A, at::Tensor B) {
   gemm_bt_8p_kernel<<<nbm * nbn, TPB8, 0, stream>>>(
       (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
       (bf16*)C.data_ptr(), M, N, K, nbm, nbn);
+  CHECK_CUDA_OK();
+  return C;
+}
+
+
+// conv3x3 fwd/dgrad on the 256x256 schedule (eligibility: channels % 64 == 0)
+at::Tensor conv3x3_8p(at::Tensor A2d, long Nn, long H, long W, long stride,
+                      at::Tensor w2d, bool dgrad) {
+  TORCH_CHECK(A2d.is_cuda() && A2d.scalar_type() == at::kBFloat16);
+  long AC = A2d.size(1);           // reduction channels (Cin fwd / Cout dgrad)
+  long NC = w2d.size(0);           // output channels
+  TORCH_CHECK(w2d.size(1) == 9 * AC && AC % 64 == 0);
+  long Hout = (H + 2 - 3) / stride + 1, Wout = (W + 2 - 3) / stride + 1;
+  long M = dgrad ? Nn * H * W : Nn * Hout * Wout;
+  auto C = at::empty({M, NC}, A2d.options());
+  Geom8 g{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
+  int nbm = (int)((M + 255) / 256), nbn = (int)((NC + 255) / 256);
+  static thread_local at::Tensor zp;
+  if (!zp.defined() || zp.device() != A2d.device())
+    zp = at::zeros({16}, A2d.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (dgrad)
+    conv3x3_8p_kernel<true><<<nbm * nbn, TPB8, 0, stream>>>(
+        (const bf16*)A2d.const_data_ptr(), (const bf16*)w2d.const_data_ptr(),
+        (bf16*)C.data_ptr(), M, (int)AC, (int)NC, g, nbm, nbn,
+        (const bf16*)zp.const_data_ptr());
+  else
+    conv3x3_8p_kernel<false><<<nbm * nbn, TPB8, 0, stream>>>(
+        (const bf16*)A2d.const_data_ptr(), (const bf16*)w2d.const_data_ptr(),
+        (bf16*)C.data_ptr(), M, (int)AC, (int)NC, g, nbm, nbn,
+        (const bf16*)zp.const_data_ptr());
   CHECK_CUDA_OK();
   return C;
 }
