@@ -141,3 +141,17 @@ def test_gemm_8p_variants_match_matmul(fn_name):
         ref = A.float() @ B.float().t()
         assert torch.allclose(C, ref, atol=0.5, rtol=0.05), \
             (C - ref).abs().max()
+
+
+def test_conv3x3_8p_matches_old():
+    """Experimental 256x256 conv3x3 stays correct (not dispatched: measured
+    at parity-to-slower vs the 128x128 kernel on ResNet shapes)."""
+    e = _ext()
+    torch.manual_seed(8)
+    n, cin, cout, hw = 4, 64, 64, 14
+    x2d = torch.randn(n * hw * hw, cin, device=DEV).bfloat16()
+    w2d = (torch.randn(cout, 9 * cin, device=DEV) * (9 * cin) ** -0.5) \
+        .bfloat16()
+    y_old = e.conv3x3_fwd(x2d, n, hw, hw, 1, w2d)
+    y_8p = e.conv3x3_8p(x2d, n, hw, hw, 1, w2d, False)
+    assert torch.allclose(y_old.float(), y_8p.float(), atol=1e-2)
