@@ -91,9 +91,13 @@ def run_worker(local_rank: int, nprocs: int, args, style: StyleConfig,
             optimizer, model.named_parameters(),
             compression=style.compression, fusion_mb=args.bucket_mb)
 
+    resume_best = 0.0
     if args.resume:
         ck = load_checkpoint(args.resume, model)
         args.start_epoch = ck.get("epoch", args.start_epoch)
+        # restore the running best so the first post-resume epoch doesn't
+        # unconditionally overwrite model_best.pth.tar
+        resume_best = float(ck.get("best_acc1", 0.0))
     elif args.pretrained and os.path.isfile("checkpoint.pth.tar"):
         # no model-zoo download in this environment: --pretrained loads the
         # local checkpoint (reference loads torchvision zoo weights here)
@@ -123,7 +127,7 @@ def run_worker(local_rank: int, nprocs: int, args, style: StyleConfig,
                         criterion, state)
 
     timer = EpochTimer(args.epoch_csv or None) if rank == 0 else EpochTimer(None)
-    best_acc1 = 0.0
+    best_acc1 = resume_best
     for epoch in range(args.start_epoch, args.epochs):
         if train_sampler is not None:
             train_sampler.set_epoch(epoch)
@@ -146,8 +150,11 @@ def run_worker(local_rank: int, nprocs: int, args, style: StyleConfig,
 
 
 def _wrap_loader(loader, style: StyleConfig, state: TrainState):
-    if style.use_prefetcher and state.device.type == "cuda" \
-            and not getattr(loader, "_amdtrain_prefetched", False):
+    # a CudaPrefetcher consumes its loader iterator, so a fresh one is built
+    # for every epoch/pass (matching the reference, which constructs
+    # data_prefetcher(loader) at the top of train()/validate(),
+    # apex_distributed.py:302,357)
+    if style.use_prefetcher and state.device.type == "cuda":
         return CudaPrefetcher(loader, device=state.device,
                               dtype=state.input_dtype(),
                               channels_last=state.channels_last)

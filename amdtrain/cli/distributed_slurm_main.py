@@ -42,7 +42,10 @@ def worker(local_gpu: int, ngpus: int, node_rank: int, world_size: int,
                         job_id=os.environ.get("SLURM_JOBID"))
     if not args.epoch_csv and rank == 0:
         args.epoch_csv = "slurm_epochs.csv"
-    run_worker(local_gpu, world_size, args, STYLE, global_rank=rank)
+    # per-GPU batch = total / ngpus_per_node (NOT / world_size) — reference
+    # distributed_slurm_main.py:155 divides by ngpus_per_node, so each node
+    # processes the full --batch-size
+    run_worker(local_gpu, ngpus, args, STYLE, global_rank=rank)
 
 
 def main(argv=None) -> None:

@@ -23,7 +23,10 @@ def base_parser(description: str) -> argparse.ArgumentParser:
                    help="model architecture: " + " | ".join(names) +
                         " (default: resnet18)")
     p.add_argument("-j", "--workers", default=4, type=int, metavar="N",
-                   help="number of data loading workers (default: 4)")
+                   help="number of data loading workers (default: 4; honored "
+                        "— the reference exposes -j 4 but hardcodes "
+                        "num_workers=2 in its DataLoaders, "
+                        "distributed.py:178; we deliberately honor the flag)")
     p.add_argument("--epochs", default=90, type=int, metavar="N",
                    help="number of total epochs to run")
     p.add_argument("--start-epoch", default=0, type=int, metavar="N",

@@ -4,7 +4,7 @@ MI355X-native equivalent of the reference's apex ``data_prefetcher``
 (apex_distributed.py:115-169): while the model computes on batch i, batch
 i+1's pinned uint8 tensor is copied host->device (``non_blocking`` ==
 ``hipMemcpyAsync`` from pinned memory) on a dedicated side HIP stream, and
-the fused cast+(x-mean)/std kernel (ops/csrc/normalize.hip) runs there too —
+the fused cast+(x-mean)/std kernel (ops/csrc/elementwise.hip) runs there too —
 so the consumer stream sees a ready, normalized tensor.  Consumer-side
 ordering uses ``wait_stream`` + ``record_stream`` exactly like the
 reference's ``next()`` (apex_distributed.py:160-168).

@@ -166,12 +166,16 @@ at::Tensor scatter_rows_x2(at::Tensor src2d, long Nn, long H, long W,
 }
 
 at::Tensor normalize_u8(at::Tensor x, std::vector<double> mean,
-                        std::vector<double> std_, bool bf16_out) {
+                        std::vector<double> std_, long dtype_code) {
+  // dtype_code: 0 = fp32, 1 = bf16, 2 = fp16 (the O2-fp16 prefetcher path)
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kByte);
   TORCH_CHECK(x.dim() == 4 && x.size(1) == 3, "expects NCHW with C=3");
   auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
   long npix = x.size(0) * x.size(2) * x.size(3);
-  auto opts = x.options().dtype(bf16_out ? at::kBFloat16 : at::kFloat);
+  at::ScalarType st = dtype_code == 1   ? at::kBFloat16
+                      : dtype_code == 2 ? at::kHalf
+                                        : at::kFloat;
+  auto opts = x.options().dtype(st);
   auto y = at::empty(xc.sizes(),
                      opts.memory_format(at::MemoryFormat::ChannelsLast));
   auto stream = at::cuda::getCurrentCUDAStream();
@@ -181,11 +185,15 @@ at::Tensor normalize_u8(at::Tensor x, std::vector<double> mean,
     m[c] = (float)mean[c];
     iv[c] = 1.f / (float)std_[c];
   }
-  if (bf16_out) {
+  if (dtype_code == 1) {
     normalize_u8_kernel<__hip_bfloat16><<<grid, AMD_TPB, 0, stream>>>(
         (const unsigned char*)xc.const_data_ptr(),
         (__hip_bfloat16*)y.data_ptr(), m[0], m[1], m[2], iv[0], iv[1], iv[2],
         npix);
+  } else if (dtype_code == 2) {
+    normalize_u8_kernel<__half><<<grid, AMD_TPB, 0, stream>>>(
+        (const unsigned char*)xc.const_data_ptr(), (__half*)y.data_ptr(),
+        m[0], m[1], m[2], iv[0], iv[1], iv[2], npix);
   } else {
     normalize_u8_kernel<float><<<grid, AMD_TPB, 0, stream>>>(
         (const unsigned char*)xc.const_data_ptr(), y.data_ptr<float>(), m[0],

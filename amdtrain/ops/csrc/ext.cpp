@@ -18,7 +18,7 @@ at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor target,
 
 // elementwise.hip
 at::Tensor normalize_u8(at::Tensor x, std::vector<double> mean,
-                        std::vector<double> std_, bool bf16_out);
+                        std::vector<double> std_, long dtype_code);
 at::Tensor topk_ranks(at::Tensor logits, at::Tensor target);
 at::Tensor scatter_rows_x2(at::Tensor src2d, long Nn, long H, long W,
                            long Hs, long Ws);

@@ -104,8 +104,8 @@ def normalize_u8(x: torch.Tensor, dtype: torch.dtype = torch.float32,
     assert x.dtype == torch.uint8 and x.dim() == 4 and x.size(1) == len(mean)
     if use_ext_for(x):
         xc = x.contiguous(memory_format=torch.channels_last)
-        return require_ext().normalize_u8(xc, list(mean), list(std),
-                                          dtype == torch.bfloat16)
+        code = {torch.float32: 0, torch.bfloat16: 1, torch.float16: 2}[dtype]
+        return require_ext().normalize_u8(xc, list(mean), list(std), code)
     m = torch.tensor(mean, dtype=torch.float32, device=x.device).reshape(1, -1, 1, 1)
     s = torch.tensor(std, dtype=torch.float32, device=x.device).reshape(1, -1, 1, 1)
     return ((x.float() - m) / s).to(dtype)

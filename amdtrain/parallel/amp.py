@@ -12,7 +12,7 @@ Two opt levels, mirroring Apex's:
   * O2 — half model + fp32 master weights: parameters are cast to the half
     dtype (BatchNorm kept fp32), the optimizer steps fp32 master copies, and
     the post-step master->model cast plus the grad unscale + inf/nan check
-    run as fused multi-tensor HIP kernels (ops/csrc/multi_tensor.hip).
+    run as fused multi-tensor HIP kernels (ops/csrc/elementwise.hip).
 
 Dynamic loss scaling matches Apex defaults: init 2**16, backoff x0.5 on
 inf/nan (step skipped), growth x2 every 2000 clean steps.
@@ -66,6 +66,10 @@ class _AmpHandle:
         # O2 master-weight state
         self.model_params: List[torch.nn.Parameter] = []
         self.master_params: List[torch.Tensor] = []
+        # set by scale_loss when it has already copied half grads into the
+        # fp32 masters (so the unscale runs at fp32); step() then skips the
+        # cast it would otherwise do
+        self.masters_have_grads = False
 
     @torch.no_grad()
     def resync_masters(self) -> None:
@@ -156,19 +160,12 @@ def _wrap_step(optimizer, handle: _AmpHandle) -> None:
         if handle.found_inf:
             handle.steps_skipped += 1
             handle.found_inf = False
+            handle.masters_have_grads = False
             return None  # Apex behavior: skip the step after overflow
         if handle.opt_level == "O2" and handle.master_params:
-            # model half grads -> master fp32 grads
-            src, dst = [], []
-            for mp, hp in zip(handle.master_params, handle.model_params):
-                if hp.grad is None:
-                    mp.grad = None
-                    continue
-                if mp.grad is None:
-                    mp.grad = torch.empty_like(mp)
-                src.append(hp.grad)
-                dst.append(mp.grad)
-            OF.multi_tensor_cast(src, dst)
+            if not handle.masters_have_grads:
+                _populate_master_grads(handle)
+            handle.masters_have_grads = False
             out = inner_step(closure, **kw) if closure is not None else inner_step(**kw)
             # master fp32 -> model half
             with torch.no_grad():
@@ -179,22 +176,49 @@ def _wrap_step(optimizer, handle: _AmpHandle) -> None:
     optimizer.step = step
 
 
+def _populate_master_grads(handle: _AmpHandle) -> None:
+    """Copy the model's half grads into the fp32 master grads (one fused
+    multi-tensor cast).  Unscaling then happens at fp32 precision, so
+    dividing by 2**16 cannot flush small fp16 gradients to zero."""
+    src, dst = [], []
+    for mp, hp in zip(handle.master_params, handle.model_params):
+        if hp.grad is None:
+            mp.grad = None
+            continue
+        if mp.grad is None:
+            mp.grad = torch.empty_like(mp)
+        src.append(hp.grad)
+        dst.append(mp.grad)
+    OF.multi_tensor_cast(src, dst)
+
+
 @contextlib.contextmanager
 def scale_loss(loss: torch.Tensor, optimizer):
     """``with amp.scale_loss(loss, optimizer) as scaled: scaled.backward()``
     (apex_distributed.py:328-329).  Scales the loss up before backward,
     unscales gradients + checks inf/nan after, and updates the dynamic scale.
+
+    In the O2 path the half model grads are first copied into the fp32
+    master grads and the unscale + inf check runs on those (Apex O2
+    semantics — unscaling in fp16 would reintroduce the underflow loss
+    scaling exists to prevent); step() consumes the populated masters.
     """
     handle: _AmpHandle = optimizer._amp_handle
     scaler = handle.scaler
     yield loss * scaler.scale if scaler.scale != 1.0 else loss
 
     # after backward: unscale grads in-place + detect overflow
-    grads = [p.grad for g in optimizer.param_groups for p in g["params"]
-             if p.grad is not None]
-    if handle.opt_level == "O2":
-        grads = [hp.grad for hp in handle.model_params if hp.grad is not None] \
-            or grads
+    if handle.opt_level == "O2" and handle.master_params:
+        _populate_master_grads(handle)
+        handle.masters_have_grads = True
+        grads = [mp.grad for mp in handle.master_params if mp.grad is not None]
+        # any non-master (fp32, e.g. BatchNorm) params still need unscaling
+        masters = set(id(mp) for mp in handle.master_params)
+        grads += [p.grad for g in optimizer.param_groups for p in g["params"]
+                  if p.grad is not None and id(p) not in masters]
+    else:
+        grads = [p.grad for g in optimizer.param_groups for p in g["params"]
+                 if p.grad is not None]
     if not grads:
         handle.found_inf = False
         return

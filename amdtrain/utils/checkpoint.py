@@ -48,7 +48,12 @@ def load_checkpoint(path: str, model=None,
     """Load a reference-schema checkpoint; optionally restore into ``model``."""
     if not os.path.isfile(path):
         raise FileNotFoundError(f"no checkpoint found at '{path}'")
-    state = torch.load(path, map_location=map_location, weights_only=False)
+    # the schema is plain tensors/scalars: prefer the pickle-safe load and
+    # only fall back for externally produced checkpoints with custom classes
+    try:
+        state = torch.load(path, map_location=map_location, weights_only=True)
+    except Exception:
+        state = torch.load(path, map_location=map_location, weights_only=False)
     if model is not None:
         module = getattr(model, "module", model)
         sd = state["state_dict"]

@@ -6,7 +6,7 @@ requested k, each ``100 * (#samples whose top-k predictions contain the
 label) / batch``.
 
 On MI355X the [B,1000] top-5 selection runs as a hand-written HIP kernel
-(amdtrain/ops/csrc/accuracy.hip) — this module dispatches to it when the
+(amdtrain/ops/csrc/elementwise.hip) — this module dispatches to it when the
 extension is loaded and the input lives on the GPU.
 """
 

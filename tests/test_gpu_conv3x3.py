@@ -53,7 +53,8 @@ def test_conv3x3_fwd_bwd(stride, cin, cout, hw):
 
 
 def test_resnet50_all_custom_convs_step():
-    """Flagship step with BOTH custom conv paths on (stem stays MIOpen)."""
+    """Flagship step with BOTH custom conv paths on (stem is the
+    conv_generic implicit-GEMM kernel, conv_stem.hip)."""
     from amdtrain.models import build_model
     from amdtrain.ops import CrossEntropyLoss, FusedSGD
     _ext()
@@ -73,7 +74,8 @@ def test_resnet50_all_custom_convs_step():
 
 
 def test_resnet18_output_parity_custom_vs_miopen(monkeypatch):
-    """Same weights, same input: custom conv stack vs MIOpen stack agree."""
+    """Same weights, same input: custom conv stack vs the torch-ROCm
+    fallback stack agree."""
     from amdtrain.models import build_model
     _ext()
     torch.manual_seed(2)
