@@ -61,6 +61,44 @@ def test_o2_master_weights_flow():
         assert torch.allclose(mp.bfloat16().float(), hp.float(), atol=1e-2)
 
 
+def test_o2_fp16_unscales_in_fp32():
+    """O2+fp16: scale_loss copies half grads into the fp32 masters BEFORE
+    unscaling, so values below fp16's normal range survive the divide
+    (the Apex O2 contract; advisor finding r1)."""
+
+    class Tiny(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.w = torch.nn.Parameter(torch.ones(4))
+
+        def forward(self, x):
+            return (self.w * x).sum()
+
+    m = Tiny()
+    opt = torch.optim.SGD(m.parameters(), lr=1.0)
+    m2, opt2 = amp.initialize(m, opt, opt_level="O2", dtype=torch.float16,
+                              init_scale=1024.0)
+    h = opt2._amp_handle
+    # true grad = 1e-6 per element: scaled by 1024 it is ~1e-3 (representable
+    # in fp16); unscaled at fp16 it would flush toward zero (min normal
+    # 6.1e-5), at fp32 it survives.
+    x = torch.full((4,), 1e-6, dtype=torch.float16)
+    loss = m2(x)
+    with amp.scale_loss(loss, opt2) as scaled:
+        scaled.backward()
+    assert h.masters_have_grads
+    mp = h.master_params[0]
+    assert mp.grad is not None and mp.grad.dtype == torch.float32
+    assert torch.allclose(mp.grad, torch.full((4,), 1e-6), rtol=0.05)
+    # model half grad is still scaled (unscale ran on the masters)
+    assert m2.w.grad.float().abs().max() > 1e-4
+    w_before = mp.detach().clone()
+    opt2.step()
+    assert not h.masters_have_grads  # consumed
+    applied = w_before - mp.detach()  # lr=1.0 -> exactly the unscaled grad
+    assert torch.allclose(applied, torch.full((4,), 1e-6), rtol=0.05)
+
+
 def test_overflow_skips_step():
     m, opt = _model_opt()
     m2, opt2 = amp.initialize(m, opt, opt_level="O1", dtype=torch.float16)

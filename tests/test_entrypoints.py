@@ -75,3 +75,49 @@ def test_resume(tmp_path):
     _run_in(tmp_path, main, COMMON)
     # resume from the checkpoint written above (start_epoch advances to 1)
     _run_in(tmp_path, main, COMMON + ["--resume", "checkpoint.pth.tar"])
+
+
+def test_resume_restores_best_acc1(tmp_path):
+    """Resume must restore best_acc1 from the checkpoint so the first
+    post-resume epoch can't silently overwrite model_best.pth.tar with a
+    worse model (advisor finding r1)."""
+    from amdtrain.cli.distributed import main
+    _run_in(tmp_path, main, COMMON)
+    ck_path = tmp_path / "checkpoint.pth.tar"
+    ck = torch.load(str(ck_path), weights_only=False)
+    ck["best_acc1"] = 99.9  # a 2-step random-init run can't beat this
+    torch.save(ck, str(ck_path))
+    # sentinel "best" file: a correct resume must NOT overwrite it
+    (tmp_path / "model_best.pth.tar").write_bytes(b"sentinel-best")
+    best_bytes = (tmp_path / "model_best.pth.tar").read_bytes()
+    # later --epochs wins in argparse: run one more epoch (start_epoch=1)
+    _run_in(tmp_path, main,
+            COMMON + ["--epochs", "2", "--resume", "checkpoint.pth.tar"])
+    new_ck = torch.load(str(ck_path), weights_only=False)
+    assert new_ck["best_acc1"] == 99.9  # carried through, not reset to 0
+    # model_best untouched (no false is_best)
+    assert (tmp_path / "model_best.pth.tar").read_bytes() == best_bytes
+
+
+def test_slurm_rank_math_and_batch_division(monkeypatch):
+    """Global rank = node_rank*ngpus + gpu (reference
+    distributed_slurm_main.py:136) and per-GPU batch divides by
+    ngpus_per_node, NOT world_size (reference :155)."""
+    from amdtrain.cli import distributed_slurm_main as M
+
+    calls = {}
+    monkeypatch.setattr(M.comm, "init_from_file",
+                        lambda **kw: calls.update(init=kw))
+
+    def fake_run_worker(local_gpu, nprocs, args, style, global_rank=None):
+        calls.update(local_gpu=local_gpu, nprocs=nprocs,
+                     global_rank=global_rank)
+    monkeypatch.setattr(M, "run_worker", fake_run_worker)
+
+    args = M.parse_args(COMMON)
+    # node 1 of 2, gpu 3 of 4 -> global rank 7; batch divisor must be 4
+    M.worker(local_gpu=3, ngpus=4, node_rank=1, world_size=8, args=args)
+    assert calls["init"]["rank"] == 7
+    assert calls["init"]["world_size"] == 8
+    assert calls["global_rank"] == 7
+    assert calls["nprocs"] == 4  # run_worker divides batch by this

@@ -34,6 +34,9 @@ def test_normalize_u8():
     yb = OF.normalize_u8(x, dtype=torch.bfloat16)
     assert yb.dtype == torch.bfloat16
     assert torch.allclose(yb.float(), ref, atol=0.05, rtol=0.02)
+    yh = OF.normalize_u8(x, dtype=torch.float16)  # O2-fp16 prefetcher path
+    assert yh.dtype == torch.float16
+    assert torch.allclose(yh.float(), ref, atol=0.01, rtol=0.005)
 
 
 # ---------- cross entropy ----------
