@@ -31,6 +31,13 @@ def _fuse_stats_enabled() -> bool:
     return os.environ.get("AMDTRAIN_FUSE_BNSTATS", "1") == "1"
 
 
+def _wgrad2_enabled() -> bool:
+    """tn2_wgrad (wgrad.hip): tr-read TN core — single launch for all 9
+    conv3x3 taps, no atomics, bitwise deterministic.  Default on;
+    AMDTRAIN_WGRAD2=0 falls back to the round-1 per-tap kernels for A/B."""
+    return os.environ.get("AMDTRAIN_WGRAD2", "1") == "1"
+
+
 class _Conv1x1(torch.autograd.Function):
     """1x1 conv; forward also emits per-block BN-statistics partials
     (non-differentiable 2nd output) for the fused conv->BN pipeline when
@@ -81,10 +88,13 @@ class _Conv1x1(torch.autograd.Function):
         if stride > 1:
             # fused zero+scatter (one write pass; stride-2 only in ResNet)
             dx = e.scatter_rows_x2(dx2d, n, h, w, ho, wo)
-            dw = e.gemm_tn_strided(gy2d, x2d, n, h, w, stride)
+            dw = e.tn2_wgrad(gy2d, x2d, 1, n, h, w, stride, 1) \
+                if _wgrad2_enabled() \
+                else e.gemm_tn_strided(gy2d, x2d, n, h, w, stride)
         else:
             dx = dx2d.view(n, ho, wo, cin).permute(0, 3, 1, 2)
-            dw = e.gemm_tn(gy2d, x2d, 0)
+            dw = e.tn2_wgrad(gy2d, x2d) if _wgrad2_enabled() \
+                else e.gemm_tn(gy2d, x2d, 0)
         return dx, dw.reshape(cout, cin, 1, 1), None
 
 
@@ -133,7 +143,10 @@ class _Conv3x3(torch.autograd.Function):
         gy2d = _rows(gy).to(torch.bfloat16)
         dx2d = e.conv3x3_dgrad(gy2d, n, h, w, stride, w2d)
         dx = dx2d.view(n, h, w, cin).permute(0, 3, 1, 2)
-        dw2d = e.conv3x3_wgrad(gy2d, x2d, n, h, w, stride)  # [Cout, 9*Cin] f32
+        # [Cout, 9*Cin] f32 — single-launch tap-gather TN (wgrad.hip)
+        dw2d = e.tn2_wgrad(gy2d, x2d, 9, n, h, w, stride, 2) \
+            if _wgrad2_enabled() \
+            else e.conv3x3_wgrad(gy2d, x2d, n, h, w, stride)
         # back to [Cout,Cin,3,3] (channels_last layout of the weight)
         dw = dw2d.view(cout, 3, 3, cin).permute(0, 3, 1, 2) \
             .contiguous(memory_format=torch.channels_last)

@@ -72,6 +72,11 @@ at::Tensor conv3x3_dgrad(at::Tensor dy2d, long Nn, long H, long W,
 at::Tensor conv3x3_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn, long H,
                          long W, long stride);
 
+// wgrad.hip
+at::Tensor tn2_wgrad(at::Tensor dY, at::Tensor X, long taps, long Nn, long H,
+                     long W, long stride, long gmode);
+at::Tensor tr16_probe(at::Tensor in);
+
 // conv_stem.hip
 at::Tensor conv_generic_fwd(at::Tensor x2d, long Nn, long H, long W, long KH,
                             long KW, long stride, long pad, at::Tensor w2);
@@ -125,6 +130,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_8p", &conv3x3_8p);
   m.def("conv3x3_dgrad", &conv3x3_dgrad);
   m.def("conv3x3_wgrad", &conv3x3_wgrad);
+  m.def("tn2_wgrad", &tn2_wgrad, py::arg("dY"), py::arg("X"),
+        py::arg("taps") = 1, py::arg("Nn") = 0, py::arg("H") = 0,
+        py::arg("W") = 0, py::arg("stride") = 1, py::arg("gmode") = 0);
+  m.def("tr16_probe", &tr16_probe);
   m.def("conv_generic_fwd", &conv_generic_fwd);
   m.def("conv_generic_wgrad", &conv_generic_wgrad);
   m.def("max_pool_3x3_s2_fwd", &max_pool_3x3_s2_fwd);

@@ -1,0 +1,406 @@
+// TN ("wgrad") GEMM core v2 for gfx950 — dW[N, K] += sum_m dY[m,n] * Xg[m,k]
+// covering conv1x1 wgrad (K = Cin), strided-1x1 wgrad (gathered X rows) and
+// conv3x3 wgrad (K = 9*Cin, im2col column gather) in ONE kernel family
+// (SURVEY §2c rows "Linear/GEMM (cuBLAS)" wgrad + "Conv2d 3x3" wgrad;
+// reference hot path /root/reference/distributed.py:268 loss.backward()).
+//
+// Why v2 (round-1 profile: conv3x3_wgrad 19 ms/step + gemm_tn 10 ms/step):
+//  * the round-1 TN kernels transposed through REGISTERS with per-element
+//    swizzled ds_write_b16 (32 x 2B writes per thread per 32-m chunk) and
+//    drained two barriers per chunk for only 16 MFMA/wave;
+//  * conv3x3 wgrad launched 9 separate kernels re-reading dY and X per tap;
+//  * split-M accumulated through fp32 atomicAdd — heavily contended on the
+//    small early-layer dW tiles.
+// v2 fixes all three:
+//  * staging is __builtin_amdgcn_global_load_lds DIRECT into a subtiled
+//    layout consumed by ds_read_b64_tr_b16 (gfx950 hardware transpose read,
+//    guide T10): no register round-trip, vectorized 16 B stores, and the
+//    9 taps are just columns of one [N, 9*Cin] output (dY staged ONCE per
+//    m-chunk instead of 9 times);
+//  * 2-phase double-buffer (guide §5.5 minimum-2-phase recipe): stage the
+//    next m-chunk while MFMAs consume the current one, one vmcnt(0)+barrier
+//    per chunk, 32 MFMA/wave between barriers (MC=64);
+//  * split-M partial outputs go to a [msplit, N*K] fp32 buffer with plain
+//    stores + a vectorized collapse kernel — no atomics, and the reduction
+//    order is FIXED, so wgrad v2 is bitwise deterministic by construction
+//    (no AMDTRAIN_DETERMINISTIC slow path needed).
+//
+// LDS layout ("tr-subtile", derived from the guide's m156 mapping): the
+// [MC x NCOLS] operand tile is stored as 256-element (512 B) groups
+// G = ((m>>5)*NCOLS/16 + (c>>4))*2 + ((m>>2)&1), element (m,c) at byte
+//   G*512 + ((m>>3)&3)*128 + (m&3)*32 + (c&15)*2 .
+// A group's four 128 B windows are exactly what one ds_read_b64_tr_b16
+// returns per 16-lane quarter (lane l, elem j <- window byte
+// (l&15)*2 + j*32), so two tr-reads (r = 0/1 groups, +512 B apart)
+// assemble the full bf16x8 MFMA operand
+//   frag[jj] = T[m = 8*(l>>4) + jj][c = l&15],   jj = 0..7.
+// Bank behavior: a staging wave's 64 concurrent 16 B stores land on the 8
+// 16 B positions of the 128 B bank cycle exactly 8x each (position =
+// (2*(m&3) + ((c>>3)&1)) mod 8) — evenly distributed, i.e. the 8-cycle
+// floor for 1 KiB of LDS writes with no aliasing penalty; each tr-read
+// window is 128 B contiguous, so reads are conflict-free too.
+#include "common.h"
+
+namespace {
+
+using bf16 = __hip_bfloat16;
+using bf16x8 = __attribute__((ext_vector_type(8))) short;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+using u32x2 = __attribute__((ext_vector_type(2))) unsigned int;
+
+constexpr int TN2_TPB = 256;  // 4 waves
+constexpr int GROUP_BYTES = 512;
+
+__device__ __forceinline__ int xcd_swz_tn2(int bid, int nwg) {
+  constexpr int NXCD = 8;
+  if (nwg < NXCD) return bid;
+  int xcd = bid % NXCD, idx = bid / NXCD;
+  int q = nwg / NXCD, r = nwg % NXCD;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+}
+
+struct TnGeom {
+  int H, W, Hout, Wout, stride;
+};
+
+// conv3x3 forward gather: output row m=(n,ho,wo) + tap -> X row or -1 (pad)
+__device__ __forceinline__ long tn2_gather3x3(long m, int kh, int kw,
+                                              const TnGeom& g) {
+  long t = m;
+  const int wo = (int)(t % g.Wout); t /= g.Wout;
+  const int ho = (int)(t % g.Hout); t /= g.Hout;
+  const int h = ho * g.stride - 1 + kh;
+  const int w = wo * g.stride - 1 + kw;
+  if (h < 0 || h >= g.H || w < 0 || w >= g.W) return -1;
+  return (t * g.H + h) * g.W + w;
+}
+
+// strided-1x1 gather: compact output row m=(n,ho,wo) -> input row
+__device__ __forceinline__ long tn2_gather_stride(long m, const TnGeom& g) {
+  long t = m;
+  const int wo = (int)(t % g.Wout); t /= g.Wout;
+  const int ho = (int)(t % g.Hout); t /= g.Hout;
+  return (t * g.H + (long)ho * g.stride) * g.W + (long)wo * g.stride;
+}
+
+// byte offset of element (m, c) of a [MC x NCOLS] tile in tr-subtile layout
+template <int NCOLS>
+__device__ __forceinline__ int st_byte(int m, int c) {
+  const int sub = (m >> 5) * (NCOLS / 16) + (c >> 4);
+  const int G = sub * 2 + ((m >> 2) & 1);
+  return G * GROUP_BYTES + ((m >> 3) & 3) * 128 + (m & 3) * 32 + (c & 15) * 2;
+}
+
+template <int NCOLS, int MC>
+constexpr int tile_bytes() {
+  return (MC / 32) * (NCOLS / 16) * 2 * GROUP_BYTES;
+}
+
+// Stage one [MC x NCOLS] operand tile into tr-subtile LDS via
+// global_load_lds.  MODE 0: rows are m (plain; also the dY operand),
+// MODE 1: strided-1x1 gather, MODE 2: conv3x3 tap gather (cols span taps).
+// Out-of-range rows/cols pull from the 16 B zero page.
+template <int MODE, int NCOLS, int MC>
+__device__ __forceinline__ void stage_tn2(
+    const bf16* __restrict__ g, int ld, long m0, long M, int col0,
+    int total_cols, int cin, const TnGeom& geo, const bf16* __restrict__ zp,
+    bf16* lds) {
+  const int t = threadIdx.x;
+  constexpr int UNITS = MC * NCOLS / 8;
+#pragma unroll
+  for (int rnd = 0; rnd < UNITS / TN2_TPB; ++rnd) {
+    const int u = rnd * TN2_TPB + t;
+    const int mloc = u / (NCOLS / 8);
+    const int c0 = (u % (NCOLS / 8)) * 8;
+    const long m = m0 + mloc;
+    const int col = col0 + c0;
+    const bf16* src = zp;
+    if (m < M && col + 8 <= total_cols) {
+      if (MODE == 0) {
+        src = g + m * (long)ld + col;
+      } else if (MODE == 1) {
+        src = g + tn2_gather_stride(m, geo) * (long)ld + col;
+      } else {
+        const int tap = col / cin;  // unit never spans taps (cin % 8 == 0)
+        const long row = tn2_gather3x3(m, tap / 3, tap % 3, geo);
+        if (row >= 0) src = g + row * (long)ld + (col - tap * cin);
+      }
+    }
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)(
+            (char*)lds + st_byte<NCOLS>(mloc, c0)),
+        16, 0, 0);
+  }
+}
+
+// Two hardware transpose reads -> one bf16x8 MFMA operand.
+// addr = LDS tile base + lane*8 + (runtime wave column offset); OFF is the
+// compile-time byte offset of the (sub, r=0) group; r=1 is the next group.
+template <int OFF>
+__device__ __forceinline__ bf16x8 tr_frag(unsigned addr) {
+  u32x2 v0, v1;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2 offset:%3\n\t"
+      "ds_read_b64_tr_b16 %1, %2 offset:%4"
+      : "=&v"(v0), "=&v"(v1)
+      : "v"(addr), "i"(OFF), "i"(OFF + GROUP_BYTES));
+  union {
+    unsigned int u[4];
+    bf16x8 f;
+  } pack;
+  pack.u[0] = v0.x;
+  pack.u[1] = v0.y;
+  pack.u[2] = v1.x;
+  pack.u[3] = v1.y;
+  return pack.f;
+}
+
+// NW x KW = wave grid over the (n, k) output tile (NW*KW == 4); per-wave
+// sub-tile is 64x64 (4x4 fragments).  MC = m rows staged per chunk.
+template <int GMODE, int NW, int KW, int MC>
+__global__ void __launch_bounds__(TN2_TPB, 2)
+tn2_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
+           float* __restrict__ parts, long M, int N, int K9, int Cin,
+           TnGeom geo, int nbn, int nbk, int msplit,
+           const bf16* __restrict__ zp) {
+  constexpr int NCY = NW * 64;   // dY tile cols
+  constexpr int NCX = KW * 64;   // X tile cols
+  constexpr int KSTEPS = MC / 32;
+  __shared__ __align__(16) char Ys[2][tile_bytes<NCY, MC>()];
+  __shared__ __align__(16) char Xs[2][tile_bytes<NCX, MC>()];
+
+  const int tiles = nbn * nbk;
+  const int bid = xcd_swz_tn2(blockIdx.x, tiles * msplit);
+  const int tile = bid % tiles;
+  const int mpart = bid / tiles;
+  const int bn = tile / nbk, bk = tile % nbk;
+  const int n0 = bn * NCY, k0 = bk * NCX;
+
+  const long mchunks = (M + MC - 1) / MC;
+  const long cpp = (mchunks + msplit - 1) / msplit;
+  const long mc0 = (long)mpart * cpp;
+  const long mc1 = min(mc0 + cpp, mchunks);
+
+  const int t = threadIdx.x;
+  const int wave = t / AMD_WAVE, lane = t % AMD_WAVE;
+  const int wn = (wave / KW) * 64;  // n offset of this wave's sub-tile
+  const int wk = (wave % KW) * 64;  // k offset
+  const int fr = lane & 15, fq = lane >> 4;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  if (mc0 < mc1) {
+    // per-wave LDS base addresses (bytes; +lane*8 is the tr-read's own slot)
+    const unsigned ybase0 =
+        (unsigned)(unsigned long long)(&Ys[0][0]) + lane * 8 +
+        (wn / 16) * 2 * GROUP_BYTES;
+    const unsigned xbase0 =
+        (unsigned)(unsigned long long)(&Xs[0][0]) + lane * 8 +
+        (wk / 16) * 2 * GROUP_BYTES;
+    constexpr unsigned YB = tile_bytes<NCY, MC>();
+    constexpr unsigned XB = tile_bytes<NCX, MC>();
+
+    stage_tn2<0, NCY, MC>(dY, N, mc0 * MC, M, n0, N, Cin, geo, zp, (bf16*)Ys[0]);
+    stage_tn2<GMODE, NCX, MC>(X, Cin, mc0 * MC, M, k0, K9, Cin, geo, zp,
+                              (bf16*)Xs[0]);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    for (long mc = mc0; mc < mc1; ++mc) {
+      const int cur = (int)(mc - mc0) & 1, nxt = cur ^ 1;
+      if (mc + 1 < mc1) {
+        stage_tn2<0, NCY, MC>(dY, N, (mc + 1) * MC, M, n0, N, Cin, geo, zp,
+                              (bf16*)Ys[nxt]);
+        stage_tn2<GMODE, NCX, MC>(X, Cin, (mc + 1) * MC, M, k0, K9, Cin, geo,
+                                  zp, (bf16*)Xs[nxt]);
+      }
+      const unsigned yb = ybase0 + cur * YB;
+      const unsigned xb = xbase0 + cur * XB;
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        bf16x8 a[4], b[4];
+        // each 16-col subtile is 2 groups; the m-half (ks) advances by a
+        // full row of subtiles
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+          a[i] = tr_frag<0>(yb + (ks * (NCY / 16) + i) * 2 * GROUP_BYTES);
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          b[j] = tr_frag<0>(xb + (ks * (NCX / 16) + j) * 2 * GROUP_BYTES);
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a[i], b[j], acc[i][j], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  // plain (non-atomic) per-mpart partial tile
+  float* out = parts + (long)mpart * N * (long)K9;
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int n = n0 + wn + i * 16 + fq * 4 + r;
+        const int k = k0 + wk + j * 16 + fr;
+        if (n < N && k < K9) out[(long)n * K9 + k] = acc[i][j][r];
+      }
+}
+
+// out[i] = sum_p parts[p][i] — fixed-order, vectorized collapse
+__global__ void __launch_bounds__(AMD_TPB)
+tn2_collapse_kernel(const float* __restrict__ parts, float* __restrict__ out,
+                    int nparts, long NK) {
+  const long nv = NK / 4;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
+       i += (long)gridDim.x * blockDim.x) {
+    f32x4 s = ((const f32x4*)parts)[i];
+    for (int p = 1; p < nparts; ++p) {
+      f32x4 v = ((const f32x4*)(parts + (long)p * NK))[i];
+      s.x += v.x; s.y += v.y; s.z += v.z; s.w += v.w;
+    }
+    ((f32x4*)out)[i] = s;
+  }
+}
+
+// semantics probe for ds_read_b64_tr_b16 (GPU test asserts the m156
+// mapping: lane l, elem j  <-  lds[(l&15) + j*16 + (l>>4)*64])
+__global__ void tr16_probe_kernel(const short* __restrict__ in,
+                                  short* __restrict__ out) {
+  __shared__ short lds[256];
+  for (int i = threadIdx.x; i < 256; i += blockDim.x) lds[i] = in[i];
+  __syncthreads();
+  if (threadIdx.x < 64) {
+    const unsigned addr =
+        (unsigned)(unsigned long long)(&lds[0]) + threadIdx.x * 8;
+    u32x2 v;
+    asm volatile("ds_read_b64_tr_b16 %0, %1 offset:0"
+                 : "=&v"(v) : "v"(addr));
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    short4 s;
+    __builtin_memcpy(&s, &v, 8);
+    out[threadIdx.x * 4 + 0] = s.x;
+    out[threadIdx.x * 4 + 1] = s.y;
+    out[threadIdx.x * 4 + 2] = s.z;
+    out[threadIdx.x * 4 + 3] = s.w;
+  }
+}
+
+static at::Tensor tn2_zero_page(const at::Tensor& like) {
+  static thread_local at::Tensor zp;
+  if (!zp.defined() || zp.device() != like.device())
+    zp = at::zeros({16}, like.options().dtype(at::kBFloat16));
+  return zp;
+}
+
+template <int GMODE, int NW, int KW, int MC>
+void tn2_launch(const at::Tensor& dY, const at::Tensor& X, at::Tensor& out,
+                long M, int N, int K9, int Cin, TnGeom geo, int target_blocks,
+                hipStream_t stream) {
+  const int nbn = (N + NW * 64 - 1) / (NW * 64);
+  const int nbk = (K9 + KW * 64 - 1) / (KW * 64);
+  const long tiles = (long)nbn * nbk;
+  const long mchunks = (M + MC - 1) / MC;
+  int msplit = (int)std::max<long>(
+      1, std::min<long>(mchunks, target_blocks / tiles));
+  auto zp = tn2_zero_page(dY);
+  at::Tensor parts = out;
+  if (msplit > 1)
+    parts = at::empty({msplit, (long)N * K9}, out.options());
+  tn2_kernel<GMODE, NW, KW, MC>
+      <<<(int)(tiles * msplit), TN2_TPB, 0, stream>>>(
+          (const bf16*)dY.const_data_ptr(), (const bf16*)X.const_data_ptr(),
+          parts.data_ptr<float>(), M, N, K9, Cin, geo, nbn, nbk, msplit,
+          (const bf16*)zp.const_data_ptr());
+  CHECK_CUDA_OK();
+  if (msplit > 1) {
+    const long NK = (long)N * K9;
+    TORCH_CHECK(NK % 4 == 0);
+    tn2_collapse_kernel<<<amd_grid(NK / 4), AMD_TPB, 0, stream>>>(
+        (const float*)parts.const_data_ptr(), out.data_ptr<float>(), msplit, NK);
+    CHECK_CUDA_OK();
+  }
+}
+
+}  // namespace
+
+// Unified TN wgrad v2.  gmode: 0 = plain rows (1x1 s1 / linear), 1 =
+// strided-1x1 row gather, 2 = conv3x3 tap gather (taps*Cin columns).
+// Returns fp32 [N, taps*Cin]; bitwise deterministic (fixed split + ordered
+// collapse).
+at::Tensor tn2_wgrad(at::Tensor dY, at::Tensor X, long taps, long Nn, long H,
+                     long W, long stride, long gmode) {
+  TORCH_CHECK(dY.is_cuda() && dY.scalar_type() == at::kBFloat16 &&
+              X.scalar_type() == at::kBFloat16);
+  auto Yc = dY.contiguous();
+  auto Xc = X.contiguous();
+  const long M = Yc.size(0);
+  const int N = (int)Yc.size(1);
+  const int Cin = (int)Xc.size(1);
+  const int K9 = (int)(taps * Cin);
+  TORCH_CHECK(Cin % 8 == 0 && N % 8 == 0);
+  TnGeom geo{0, 0, 0, 0, 1};
+  if (gmode == 1) {
+    long Hout = (H + stride - 1) / stride, Wout = (W + stride - 1) / stride;
+    TORCH_CHECK(M == Nn * Hout * Wout);
+    geo = TnGeom{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
+  } else if (gmode == 2) {
+    long Hout = (H + 2 - 3) / stride + 1, Wout = (W + 2 - 3) / stride + 1;
+    TORCH_CHECK(M == Nn * Hout * Wout && taps == 9);
+    geo = TnGeom{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
+  } else {
+    TORCH_CHECK(M == Xc.size(0));
+  }
+  auto out = at::empty({(long)N, (long)K9}, Yc.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const char* tb = std::getenv("AMDTRAIN_TN2_BLOCKS");
+  const int target = tb ? atoi(tb) : 512;
+
+  // config: (2,2) 128x128 tile MC=64 default; skinny-N with wide K -> (1,4);
+  // skinny-K with tall N -> (4,1) (both MC=32 to keep 2 blocks/CU in LDS)
+  if (N < 128 && K9 >= 256) {
+    if (gmode == 2)
+      tn2_launch<2, 1, 4, 32>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
+    else if (gmode == 1)
+      tn2_launch<1, 2, 2, 64>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
+    else
+      tn2_launch<0, 1, 4, 32>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
+  } else if (K9 < 128 && N >= 256 && gmode == 0) {
+    tn2_launch<0, 4, 1, 32>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
+  } else {
+    if (gmode == 2)
+      tn2_launch<2, 2, 2, 64>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
+    else if (gmode == 1)
+      tn2_launch<1, 2, 2, 64>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
+    else
+      tn2_launch<0, 2, 2, 64>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
+  }
+  return out;
+}
+
+at::Tensor tr16_probe(at::Tensor in) {
+  TORCH_CHECK(in.is_cuda() && in.scalar_type() == at::kShort &&
+              in.numel() == 256);
+  auto inc = in.contiguous();
+  auto out = at::zeros({64, 4}, inc.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  tr16_probe_kernel<<<1, 64, 0, stream>>>(
+      (const short*)inc.const_data_ptr(), out.data_ptr<short>());
+  CHECK_CUDA_OK();
+  return out;
+}

@@ -28,6 +28,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "conv3x3.hip"),
         os.path.join(CSRC, "conv_stem.hip"),
         os.path.join(CSRC, "gemm8p.hip"),
+        os.path.join(CSRC, "wgrad.hip"),
         os.path.join(CSRC, "pool.hip"),
     ],
     extra_compile_args={
