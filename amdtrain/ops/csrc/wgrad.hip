@@ -100,20 +100,34 @@ constexpr int tile_bytes() {
 // global_load_lds.  MODE 0: rows are m (plain; also the dY operand),
 // MODE 1: strided-1x1 gather, MODE 2: conv3x3 tap gather (cols span taps).
 // Out-of-range rows/cols pull from the 16 B zero page.
+//
+// global_load_lds is a WAVE-UNIFORM-BASE DMA: the hardware writes lane i's
+// 16 B at base + i*16 regardless of per-lane LDS addresses (guide: "CDNA's
+// is wave-uniform-base not per-lane scatter").  So each WAVE stages one
+// complete 1 KiB subtile (32 m x 16 c), whose tr-layout bytes are exactly
+// subtile_base + lane*16 under the lane -> (m, c) decode
+//   m' = 8*((lane>>3)&3) + 4*(lane>>5) + ((lane>>1)&3),
+//   c  = (lane&1)*8 .. +8
+// (inverse of st_byte: lane*16 = r*512 + q*128 + j*32 + c_half*2).
 template <int MODE, int NCOLS, int MC>
 __device__ __forceinline__ void stage_tn2(
     const bf16* __restrict__ g, int ld, long m0, long M, int col0,
     int total_cols, int cin, const TnGeom& geo, const bf16* __restrict__ zp,
     bf16* lds) {
   const int t = threadIdx.x;
-  constexpr int UNITS = MC * NCOLS / 8;
+  const int wave = t / AMD_WAVE, lane = t % AMD_WAVE;
+  constexpr int SUBTILES = (MC / 32) * (NCOLS / 16);
+  const int r = lane >> 5, q = (lane >> 3) & 3, j = (lane >> 1) & 3;
+  const int c_half = (lane & 1) * 8;
+  const int mfrag = q * 8 + r * 4 + j;  // m offset within the 32-row block
 #pragma unroll
-  for (int rnd = 0; rnd < UNITS / TN2_TPB; ++rnd) {
-    const int u = rnd * TN2_TPB + t;
-    const int mloc = u / (NCOLS / 8);
-    const int c0 = (u % (NCOLS / 8)) * 8;
+  for (int ss = 0; ss < SUBTILES / 4; ++ss) {
+    const int s = ss * 4 + wave;
+    const int mb = s / (NCOLS / 16);         // 32-row block
+    const int cs = (s % (NCOLS / 16)) * 16;  // subtile column base
+    const int mloc = mb * 32 + mfrag;
     const long m = m0 + mloc;
-    const int col = col0 + c0;
+    const int col = col0 + cs + c_half;
     const bf16* src = zp;
     if (m < M && col + 8 <= total_cols) {
       if (MODE == 0) {
@@ -129,7 +143,7 @@ __device__ __forceinline__ void stage_tn2(
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)src,
         (__attribute__((address_space(3))) unsigned int*)(
-            (char*)lds + st_byte<NCOLS>(mloc, c0)),
+            (char*)lds + s * 1024 + lane * 16),
         16, 0, 0);
   }
 }
