@@ -196,9 +196,22 @@ class _ConvGeneric(torch.autograd.Function):
         n, cin, h, w, kh, kw, stride, pad, cout, k = ctx.meta
         gy2d = _rows(grad_y.contiguous(memory_format=torch.channels_last)) \
             .to(torch.bfloat16)
-        dw2 = e.conv_generic_wgrad(gy2d, x2d, n, h, w, kh, kw, stride, pad)
-        dw = dw2[:, :k].view(cout, kh, kw, cin).permute(0, 3, 1, 2) \
-            .contiguous(memory_format=torch.channels_last)
+        if _wgrad2_enabled() and cout % 8 == 0:
+            # tap-gather TN core; channels padded to a multiple of 8 so a
+            # 16 B staging unit never spans taps (the stem's Cin=3 -> 8)
+            cinp = (cin + 7) // 8 * 8
+            xp = x2d if cinp == cin else \
+                torch.nn.functional.pad(x2d, (0, cinp - cin))
+            dwp = e.tn2_wgrad(gy2d, xp, kh * kw, n, h, w, stride, 2,
+                              kh, kw, pad)                 # [Cout, taps*cinp]
+            dw = dwp.view(cout, kh * kw, cinp)[:, :, :cin] \
+                .view(cout, kh, kw, cin).permute(0, 3, 1, 2) \
+                .contiguous(memory_format=torch.channels_last)
+        else:
+            dw2 = e.conv_generic_wgrad(gy2d, x2d, n, h, w, kh, kw, stride,
+                                       pad)
+            dw = dw2[:, :k].view(cout, kh, kw, cin).permute(0, 3, 1, 2) \
+                .contiguous(memory_format=torch.channels_last)
         return None, dw, None, None
 
 

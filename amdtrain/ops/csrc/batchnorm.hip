@@ -21,14 +21,21 @@ namespace {
 
 // ---- reduction: per-channel (a, b) sums over rows -------------------------
 // fwd (BWD=false): a = x, b = x^2
-// bwd (BWD=true):  a = ghat, b = ghat * xhat,
-//                  ghat = go * (y > 0) when RELU else go
-template <typename T, int VEC, bool BWD, bool RELU>
+// bwd (BWD=true):  a = ghat, b = ghat * xhat, where
+//   MASK 0: ghat = go (no relu / pre-masked input)
+//   MASK 1: ghat = go * (y > 0)                  (block-tail BN: y has addend)
+//   MASK 2: ghat = go * (scale*x + shift > 0)    (inner BN: skip the y read)
+// WG: additionally materialize ghat (consumed by the mask-free bwd apply and
+// as the residual addend gradient) — turns reduce into 3r+1w so apply drops
+// to 2r+1w (guide: every pass here is an HBM-bound R x C stream).
+template <typename T, int VEC, bool BWD, int MASK, bool WG>
 __global__ void __launch_bounds__(AMD_TPB)
 bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
                  const T* __restrict__ y, const float* __restrict__ mean,
-                 const float* __restrict__ invstd, float* __restrict__ out,
-                 long R, int C) {
+                 const float* __restrict__ invstd,
+                 const float* __restrict__ scale,
+                 const float* __restrict__ shift, T* __restrict__ ghat_out,
+                 float* __restrict__ out, long R, int C) {
   const int t = threadIdx.x;
   const int gpr = C / VEC;  // channel-groups per row
   // rows are split across blocks
@@ -45,35 +52,42 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
     float sa[VEC], sb[VEC];
 #pragma unroll
     for (int k = 0; k < VEC; ++k) sa[k] = sb[k] = 0.f;
-    float mk[VEC], ik[VEC];
+    float mk[VEC], ik[VEC], sck[VEC], shk[VEC];
     if (BWD) {
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
         mk[k] = mean[c0 + k];
         ik[k] = invstd[c0 + k];
+        if (MASK == 2) {
+          sck[k] = scale[c0 + k];
+          shk[k] = shift[c0 + k];
+        }
       }
     }
     for (long r = r0 + phase; r < r1; r += rstep) {
       const long base = r * C + c0;
       Pack<T, VEC> xv = *(const Pack<T, VEC>*)(x + base);
-      Pack<T, VEC> gv, yv;
+      Pack<T, VEC> gv, yv, gh;
       if (BWD) {
         gv = *(const Pack<T, VEC>*)(go + base);
-        if (RELU) yv = *(const Pack<T, VEC>*)(y + base);
+        if (MASK == 1) yv = *(const Pack<T, VEC>*)(y + base);
       }
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
         float xe = to_f32(xv.v[k]);
         if (BWD) {
           float ge = to_f32(gv.v[k]);
-          if (RELU && to_f32(yv.v[k]) <= 0.f) ge = 0.f;
+          if (MASK == 1 && to_f32(yv.v[k]) <= 0.f) ge = 0.f;
+          if (MASK == 2 && sck[k] * xe + shk[k] <= 0.f) ge = 0.f;
           sa[k] += ge;
           sb[k] += ge * (xe - mk[k]) * ik[k];
+          if (WG) gh.v[k] = from_f32<T>(ge);
         } else {
           sa[k] += xe;
           sb[k] += xe * xe;
         }
       }
+      if (BWD && WG) *(Pack<T, VEC>*)(ghat_out + base) = gh;
     }
     // LDS reduce across phases, then one global atomicAdd per channel
     __shared__ float lds[AMD_TPB * VEC];
@@ -108,7 +122,7 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
     // wide-C: each thread owns up to 2 channel-groups, no cross-thread reduce
     const int cpt = gpr / AMD_TPB;  // host guarantees <= 2 and divisible
     float sa[2][VEC], sb[2][VEC];
-    float mk[2][VEC], ik[2][VEC];
+    float mk[2][VEC], ik[2][VEC], sck[2][VEC], shk[2][VEC];
 #pragma unroll
     for (int j = 0; j < 2; ++j)
 #pragma unroll
@@ -117,6 +131,10 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
         if (BWD && j < cpt) {
           mk[j][k] = mean[(t + j * AMD_TPB) * VEC + k];
           ik[j][k] = invstd[(t + j * AMD_TPB) * VEC + k];
+          if (MASK == 2) {
+            sck[j][k] = scale[(t + j * AMD_TPB) * VEC + k];
+            shk[j][k] = shift[(t + j * AMD_TPB) * VEC + k];
+          }
         }
       }
     for (long r = r0; r < r1; ++r) {
@@ -124,24 +142,27 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
         const int c0 = (t + j * AMD_TPB) * VEC;
         const long base = r * C + c0;
         Pack<T, VEC> xv = *(const Pack<T, VEC>*)(x + base);
-        Pack<T, VEC> gv, yv;
+        Pack<T, VEC> gv, yv, gh;
         if (BWD) {
           gv = *(const Pack<T, VEC>*)(go + base);
-          if (RELU) yv = *(const Pack<T, VEC>*)(y + base);
+          if (MASK == 1) yv = *(const Pack<T, VEC>*)(y + base);
         }
 #pragma unroll
         for (int k = 0; k < VEC; ++k) {
           float xe = to_f32(xv.v[k]);
           if (BWD) {
             float ge = to_f32(gv.v[k]);
-            if (RELU && to_f32(yv.v[k]) <= 0.f) ge = 0.f;
+            if (MASK == 1 && to_f32(yv.v[k]) <= 0.f) ge = 0.f;
+            if (MASK == 2 && sck[j][k] * xe + shk[j][k] <= 0.f) ge = 0.f;
             sa[j][k] += ge;
             sb[j][k] += ge * (xe - mk[j][k]) * ik[j][k];
+            if (WG) gh.v[k] = from_f32<T>(ge);
           } else {
             sa[j][k] += xe;
             sb[j][k] += xe * xe;
           }
         }
+        if (BWD && WG) *(Pack<T, VEC>*)(ghat_out + base) = gh;
       }
     }
     float* part = out + (long)blockIdx.x * 2 * C;
@@ -217,7 +238,10 @@ __global__ void bn_finalize_eval_kernel(
 }
 
 // ---- apply: y = scale*x + shift (+z), relu --------------------------------
-template <typename T, int VEC, bool RELU, bool HAS_ADD>
+// 2-way unrolled (two independent 16 B streams in flight per thread) with a
+// pow2 fast path for the channel-group decompose (all ResNet C are pow2;
+// the 64-bit % was ~25 VALU cycles per 16 B).
+template <typename T, int VEC, bool RELU, bool HAS_ADD, bool POW2>
 __global__ void __launch_bounds__(AMD_TPB)
 bn_apply_kernel(const T* __restrict__ x, const T* __restrict__ z,
                 T* __restrict__ y, const float* __restrict__ scale,
@@ -229,22 +253,34 @@ bn_apply_kernel(const T* __restrict__ x, const T* __restrict__ z,
     sm[C + c] = shift[c];
   }
   __syncthreads();
+  const long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total_vec;
-       i += (long)gridDim.x * blockDim.x) {
-    const int c0 = (int)(i % gpr) * VEC;
-    const long base = i * VEC;
-    Pack<T, VEC> xv = *(const Pack<T, VEC>*)(x + base);
-    Pack<T, VEC> zv;
-    if (HAS_ADD) zv = *(const Pack<T, VEC>*)(z + base);
-    Pack<T, VEC> yv;
+       i += 2 * stride) {
+    const bool has2 = i + stride < total_vec;
+    const long i2 = has2 ? i + stride : i;  // clamp: loads stay in bounds
+    Pack<T, VEC> xv = *(const Pack<T, VEC>*)(x + i * VEC);
+    Pack<T, VEC> xv2 = *(const Pack<T, VEC>*)(x + i2 * VEC);
+    Pack<T, VEC> zv, zv2;
+    if (HAS_ADD) {
+      zv = *(const Pack<T, VEC>*)(z + i * VEC);
+      zv2 = *(const Pack<T, VEC>*)(z + i2 * VEC);
+    }
+    const int c0 = (int)(POW2 ? (i & (gpr - 1)) : (i % gpr)) * VEC;
+    const int c02 = (int)(POW2 ? (i2 & (gpr - 1)) : (i2 % gpr)) * VEC;
+    Pack<T, VEC> yv, yv2;
 #pragma unroll
     for (int k = 0; k < VEC; ++k) {
       float v = sm[c0 + k] * to_f32(xv.v[k]) + sm[C + c0 + k];
       if (HAS_ADD) v += to_f32(zv.v[k]);
       if (RELU) v = fmaxf(v, 0.f);
       yv.v[k] = from_f32<T>(v);
+      float v2 = sm[c02 + k] * to_f32(xv2.v[k]) + sm[C + c02 + k];
+      if (HAS_ADD) v2 += to_f32(zv2.v[k]);
+      if (RELU) v2 = fmaxf(v2, 0.f);
+      yv2.v[k] = from_f32<T>(v2);
     }
-    *(Pack<T, VEC>*)(y + base) = yv;
+    *(Pack<T, VEC>*)(y + i * VEC) = yv;
+    if (has2) *(Pack<T, VEC>*)(y + i2 * VEC) = yv2;
   }
 }
 
@@ -272,41 +308,66 @@ __global__ void bn_bwd_finalize_kernel(
 }
 
 // ---- bwd apply: gx = A*ghat + B*x + D -------------------------------------
-template <typename T, int VEC, bool RELU, bool WRITE_GHAT>
+// 2-way unrolled + pow2 fast path (see bn_apply_kernel).  MASK 0: go is
+// already masked (the reduce pass wrote ghat, or no relu) — no y read at
+// all; MASK 2: relu mask recomputed from the forward's scale/shift
+// (y = relu(scale*x+shift) so y>0 <=> scale*x+shift>0) — also no y read.
+template <typename T, int VEC, int MASK, bool WRITE_GHAT, bool POW2>
 __global__ void __launch_bounds__(AMD_TPB)
 bn_bwd_apply_kernel(const T* __restrict__ x, const T* __restrict__ go,
-                    const T* __restrict__ y, T* __restrict__ gx,
+                    T* __restrict__ gx,
                     T* __restrict__ ghat_out, const float* __restrict__ A,
                     const float* __restrict__ Bc, const float* __restrict__ Dc,
-                    long total_vec, int C) {
-  extern __shared__ float sm[];  // [3][C]
+                    const float* __restrict__ scale,
+                    const float* __restrict__ shift, long total_vec, int C) {
+  extern __shared__ float sm[];  // [3][C] (+[2][C] for MASK 2)
   for (int c = threadIdx.x; c < C; c += blockDim.x) {
     sm[c] = A[c];
     sm[C + c] = Bc[c];
     sm[2 * C + c] = Dc[c];
+    if (MASK == 2) {
+      sm[3 * C + c] = scale[c];
+      sm[4 * C + c] = shift[c];
+    }
   }
   __syncthreads();
   const int gpr = C / VEC;
+  const long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total_vec;
-       i += (long)gridDim.x * blockDim.x) {
-    const int c0 = (int)(i % gpr) * VEC;
-    const long base = i * VEC;
-    Pack<T, VEC> xv = *(const Pack<T, VEC>*)(x + base);
-    Pack<T, VEC> gv = *(const Pack<T, VEC>*)(go + base);
-    Pack<T, VEC> yv;
-    if (RELU) yv = *(const Pack<T, VEC>*)(y + base);
-    Pack<T, VEC> out, gh;
+       i += 2 * stride) {
+    const bool has2 = i + stride < total_vec;
+    const long i2 = has2 ? i + stride : i;  // clamp: loads stay in bounds
+    Pack<T, VEC> xv = *(const Pack<T, VEC>*)(x + i * VEC);
+    Pack<T, VEC> gv = *(const Pack<T, VEC>*)(go + i * VEC);
+    Pack<T, VEC> xv2 = *(const Pack<T, VEC>*)(x + i2 * VEC);
+    Pack<T, VEC> gv2 = *(const Pack<T, VEC>*)(go + i2 * VEC);
+    const int c0 = (int)(POW2 ? (i & (gpr - 1)) : (i % gpr)) * VEC;
+    const int c02 = (int)(POW2 ? (i2 & (gpr - 1)) : (i2 % gpr)) * VEC;
+    Pack<T, VEC> out, gh, out2, gh2;
 #pragma unroll
     for (int k = 0; k < VEC; ++k) {
+      float xe = to_f32(xv.v[k]);
       float ge = to_f32(gv.v[k]);
-      if (RELU && to_f32(yv.v[k]) <= 0.f) ge = 0.f;
-      float v = sm[c0 + k] * ge + sm[C + c0 + k] * to_f32(xv.v[k]) +
-                sm[2 * C + c0 + k];
+      if (MASK == 2 && sm[3 * C + c0 + k] * xe + sm[4 * C + c0 + k] <= 0.f)
+        ge = 0.f;
+      float v = sm[c0 + k] * ge + sm[C + c0 + k] * xe + sm[2 * C + c0 + k];
       out.v[k] = from_f32<T>(v);
       if (WRITE_GHAT) gh.v[k] = from_f32<T>(ge);
+      float xe2 = to_f32(xv2.v[k]);
+      float ge2 = to_f32(gv2.v[k]);
+      if (MASK == 2 && sm[3 * C + c02 + k] * xe2 + sm[4 * C + c02 + k] <= 0.f)
+        ge2 = 0.f;
+      float v2 = sm[c02 + k] * ge2 + sm[C + c02 + k] * xe2 +
+                 sm[2 * C + c02 + k];
+      out2.v[k] = from_f32<T>(v2);
+      if (WRITE_GHAT) gh2.v[k] = from_f32<T>(ge2);
     }
-    *(Pack<T, VEC>*)(gx + base) = out;
-    if (WRITE_GHAT) *(Pack<T, VEC>*)(ghat_out + base) = gh;
+    *(Pack<T, VEC>*)(gx + i * VEC) = out;
+    if (WRITE_GHAT) *(Pack<T, VEC>*)(ghat_out + i * VEC) = gh;
+    if (has2) {
+      *(Pack<T, VEC>*)(gx + i2 * VEC) = out2;
+      if (WRITE_GHAT) *(Pack<T, VEC>*)(ghat_out + i2 * VEC) = gh2;
+    }
   }
 }
 
@@ -327,15 +388,20 @@ static void dispatch_vec(const at::Tensor& x, F fn) {
         using devT = typename DevT<scalar_t>::type;
         constexpr int VEC = 16 / sizeof(devT);
         TORCH_CHECK(C % VEC == 0, "C must be a multiple of ", VEC);
+        TORCH_CHECK((C & (C - 1)) == 0,
+                    "C must be a power of two (apply-kernel fast path): ", C);
         TORCH_CHECK(C / VEC <= 2 * AMD_TPB, "C too large: ", C);
         fn((devT*)nullptr, std::integral_constant<int, VEC>{});
       });
 }
 
 static int bn_reduce_grid(long R, int C) {
-  // enough blocks to saturate HBM reads; partial-buffer cost is grid*2C f32
-  long rows_per_block = std::max<long>(1, R / 512);
-  return (int)std::min<long>((R + rows_per_block - 1) / rows_per_block, 512);
+  // enough blocks to saturate HBM reads (1536 = 6 blocks/CU keeps 12
+  // waves/SIMD in flight over the 3-stream bwd reduce; 512 measured ~56%
+  // of achievable BW); partial-buffer cost is grid*2C f32, folded by
+  // bn_collapse_partials
+  long rows_per_block = std::max<long>(1, R / 1536);
+  return (int)std::min<long>((R + rows_per_block - 1) / rows_per_block, 1536);
 }
 
 }  // namespace
@@ -360,9 +426,10 @@ std::vector<at::Tensor> batch_norm_fwd_train(
   dispatch_vec(x, [&](auto* tp, auto vec) {
     using devT = std::remove_pointer_t<decltype(tp)>;
     constexpr int VEC = decltype(vec)::value;
-    bn_reduce_kernel<devT, VEC, false, false>
+    bn_reduce_kernel<devT, VEC, false, 0, false>
         <<<rgrid, AMD_TPB, 0, stream>>>((const devT*)x.const_data_ptr(),
                                         nullptr, nullptr, nullptr, nullptr,
+                                        nullptr, nullptr, nullptr,
                                         sums.data_ptr<float>(), R, (int)C);
     CHECK_CUDA_OK();
     auto sums2 = at::empty({BN_COLLAPSE, 2 * C}, opts);
@@ -387,7 +454,7 @@ std::vector<at::Tensor> batch_norm_fwd_train(
     const devT* zp =
         addend ? (const devT*)addend->const_data_ptr() : nullptr;
 #define APPLY(RELU_, ADD_)                                                  \
-  bn_apply_kernel<devT, VEC, RELU_, ADD_>                                   \
+  bn_apply_kernel<devT, VEC, RELU_, ADD_, true>                             \
       <<<agrid, AMD_TPB, smem, stream>>>(                                   \
           (const devT*)x.const_data_ptr(), zp, (devT*)y.data_ptr(),         \
           scale.data_ptr<float>(), shift.data_ptr<float>(),     \
@@ -399,7 +466,7 @@ std::vector<at::Tensor> batch_norm_fwd_train(
 #undef APPLY
     CHECK_CUDA_OK();
   });
-  return {y, mean, invstd};
+  return {y, mean, invstd, scale, shift};
 }
 
 // forward-train using per-block statistics partials produced by the conv
@@ -446,7 +513,7 @@ std::vector<at::Tensor> batch_norm_fwd_train_from_parts(
     const devT* zp =
         addend ? (const devT*)addend->const_data_ptr() : nullptr;
 #define APPLY2(RELU_, ADD_)                                                 \
-  bn_apply_kernel<devT, VEC, RELU_, ADD_>                                   \
+  bn_apply_kernel<devT, VEC, RELU_, ADD_, true>                             \
       <<<agrid, AMD_TPB, smem, stream>>>(                                   \
           (const devT*)x.const_data_ptr(), zp, (devT*)y.data_ptr(),         \
           scale.data_ptr<float>(), shift.data_ptr<float>(),                 \
@@ -458,7 +525,7 @@ std::vector<at::Tensor> batch_norm_fwd_train_from_parts(
 #undef APPLY2
     CHECK_CUDA_OK();
   });
-  return {y, mean, invstd};
+  return {y, mean, invstd, scale, shift};
 }
 
 at::Tensor batch_norm_fwd_eval(at::Tensor x, at::Tensor weight,
@@ -488,7 +555,7 @@ at::Tensor batch_norm_fwd_eval(at::Tensor x, at::Tensor weight,
     const devT* zp =
         addend ? (const devT*)addend->const_data_ptr() : nullptr;
 #define APPLY(RELU_, ADD_)                                                  \
-  bn_apply_kernel<devT, VEC, RELU_, ADD_>                                   \
+  bn_apply_kernel<devT, VEC, RELU_, ADD_, true>                             \
       <<<agrid, AMD_TPB, smem, stream>>>(                                   \
           (const devT*)x.const_data_ptr(), zp, (devT*)y.data_ptr(),         \
           scale.data_ptr<float>(), shift.data_ptr<float>(),     \
@@ -506,7 +573,19 @@ at::Tensor batch_norm_fwd_eval(at::Tensor x, at::Tensor weight,
 std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
                                        at::Tensor y, at::Tensor weight,
                                        at::Tensor mean, at::Tensor invstd,
-                                       bool relu, bool need_ghat) {
+                                       bool relu, bool need_ghat,
+                                       std::optional<at::Tensor> scale,
+                                       std::optional<at::Tensor> shift) {
+  // Pass structure (every kernel is an HBM-bound R x C stream, so passes
+  // are the whole cost):
+  //  * no relu:                reduce(x,go) -> apply(x,go -> gx)
+  //  * relu + addend (bn3):    reduce(x,go,y -> ghat) [mask from y]
+  //                            -> apply(x,ghat -> gx); ghat returned as the
+  //                            residual addend gradient (1 pass saved vs
+  //                            masking again + writing ghat in apply)
+  //  * relu, no addend (bn1/2) with fwd scale/shift: mask recomputed as
+  //                            scale*x+shift>0 in BOTH passes — the y read
+  //                            disappears entirely (2 passes saved)
   check_nhwc(x);
   check_nhwc(grad_out);
   const long C = x.size(1);
@@ -520,21 +599,29 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
   auto Bc = at::empty({C}, opts);
   auto Dc = at::empty({C}, opts);
   auto gx = at::empty_like(x);
+  const bool affine_mask = relu && !need_ghat && scale && shift;
+  const bool mask_y = relu && !affine_mask;  // reduce masks from y + writes ghat
   at::Tensor ghat;
-  if (relu && need_ghat) ghat = at::empty_like(grad_out);
+  if (mask_y) ghat = at::empty_like(grad_out);
   auto stream = at::cuda::getCurrentCUDAStream();
 
   dispatch_vec(x, [&](auto* tp, auto vec) {
     using devT = std::remove_pointer_t<decltype(tp)>;
     constexpr int VEC = decltype(vec)::value;
-#define REDUCE(RELU_)                                                       \
-  bn_reduce_kernel<devT, VEC, true, RELU_><<<rgrid, AMD_TPB, 0, stream>>>(  \
-      (const devT*)x.const_data_ptr(),                                      \
-      (const devT*)grad_out.const_data_ptr(),                               \
-      (const devT*)y.const_data_ptr(), mean.data_ptr<float>(),        \
-      invstd.data_ptr<float>(), sums.data_ptr<float>(), R, (int)C)
-    if (relu) REDUCE(true);
-    else REDUCE(false);
+    const float* scp = affine_mask ? scale->data_ptr<float>() : nullptr;
+    const float* shp = affine_mask ? shift->data_ptr<float>() : nullptr;
+#define REDUCE(MASK_, WG_)                                                  \
+  bn_reduce_kernel<devT, VEC, true, MASK_, WG_>                             \
+      <<<rgrid, AMD_TPB, 0, stream>>>(                                      \
+          (const devT*)x.const_data_ptr(),                                  \
+          (const devT*)grad_out.const_data_ptr(),                           \
+          (const devT*)y.const_data_ptr(), mean.data_ptr<float>(),          \
+          invstd.data_ptr<float>(), scp, shp,                               \
+          WG_ ? (devT*)ghat.data_ptr() : nullptr,                           \
+          sums.data_ptr<float>(), R, (int)C)
+    if (mask_y) REDUCE(1, true);
+    else if (affine_mask) REDUCE(2, false);
+    else REDUCE(0, false);
 #undef REDUCE
     CHECK_CUDA_OK();
     auto sums2 = at::empty({BN_COLLAPSE, 2 * C}, opts);
@@ -554,22 +641,20 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
     CHECK_CUDA_OK();
     long total_vec = R * C / VEC;
     int agrid = amd_grid(total_vec);
-    size_t smem = 3 * C * sizeof(float);
-#define BAPPLY(RELU_, WG_)                                                  \
-  bn_bwd_apply_kernel<devT, VEC, RELU_, WG_>                                \
+    size_t smem = (affine_mask ? 5 : 3) * C * sizeof(float);
+    const devT* go_in = mask_y ? (const devT*)ghat.const_data_ptr()
+                               : (const devT*)grad_out.const_data_ptr();
+#define BAPPLY(MASK_)                                                       \
+  bn_bwd_apply_kernel<devT, VEC, MASK_, false, true>                        \
       <<<agrid, AMD_TPB, smem, stream>>>(                                   \
-          (const devT*)x.const_data_ptr(),                                  \
-          (const devT*)grad_out.const_data_ptr(),                           \
-          (const devT*)y.const_data_ptr(), (devT*)gx.data_ptr(),            \
-          WG_ ? (devT*)ghat.data_ptr() : nullptr,                         \
-          A.data_ptr<float>(), Bc.data_ptr<float>(),            \
-          Dc.data_ptr<float>(), total_vec, (int)C)
-    if (relu && need_ghat) BAPPLY(true, true);
-    else if (relu) BAPPLY(true, false);
-    else BAPPLY(false, false);
+          (const devT*)x.const_data_ptr(), go_in, (devT*)gx.data_ptr(),     \
+          nullptr, A.data_ptr<float>(), Bc.data_ptr<float>(),               \
+          Dc.data_ptr<float>(), scp, shp, total_vec, (int)C)
+    if (affine_mask) BAPPLY(2);
+    else BAPPLY(0);
 #undef BAPPLY
     CHECK_CUDA_OK();
   });
-  if (!(relu && need_ghat)) ghat = grad_out;  // placeholder / identity
+  if (!mask_y) ghat = grad_out;  // placeholder / identity (no addend)
   return {gx, gw, gb, ghat};
 }

@@ -43,7 +43,9 @@ at::Tensor batch_norm_fwd_eval(at::Tensor x, at::Tensor weight,
 std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
                                        at::Tensor y, at::Tensor weight,
                                        at::Tensor mean, at::Tensor invstd,
-                                       bool relu, bool need_ghat);
+                                       bool relu, bool need_ghat,
+                                       std::optional<at::Tensor> scale,
+                                       std::optional<at::Tensor> shift);
 
 // gemm.hip
 at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out);
@@ -74,7 +76,8 @@ at::Tensor conv3x3_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn, long H,
 
 // wgrad.hip
 at::Tensor tn2_wgrad(at::Tensor dY, at::Tensor X, long taps, long Nn, long H,
-                     long W, long stride, long gmode);
+                     long W, long stride, long gmode, long kh, long kw,
+                     long pad);
 at::Tensor tr16_probe(at::Tensor in);
 
 // conv_stem.hip
@@ -114,7 +117,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("weight"), py::arg("bias"),
         py::arg("running_mean"), py::arg("running_var"), py::arg("eps"),
         py::arg("relu"), py::arg("addend") = std::nullopt);
-  m.def("batch_norm_bwd", &batch_norm_bwd);
+  m.def("batch_norm_bwd", &batch_norm_bwd, py::arg("x"), py::arg("grad_out"),
+        py::arg("y"), py::arg("weight"), py::arg("mean"), py::arg("invstd"),
+        py::arg("relu"), py::arg("need_ghat"),
+        py::arg("scale") = std::nullopt, py::arg("shift") = std::nullopt);
   m.def("gemm_bt", &gemm_bt, py::arg("A"), py::arg("B"),
         py::arg("f32_out") = false);
   m.def("gemm_tn", &gemm_tn, py::arg("dY"), py::arg("X"),
@@ -132,7 +138,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_wgrad", &conv3x3_wgrad);
   m.def("tn2_wgrad", &tn2_wgrad, py::arg("dY"), py::arg("X"),
         py::arg("taps") = 1, py::arg("Nn") = 0, py::arg("H") = 0,
-        py::arg("W") = 0, py::arg("stride") = 1, py::arg("gmode") = 0);
+        py::arg("W") = 0, py::arg("stride") = 1, py::arg("gmode") = 0,
+        py::arg("kh") = 3, py::arg("kw") = 3, py::arg("pad") = 1);
   m.def("tr16_probe", &tr16_probe);
   m.def("conv_generic_fwd", &conv_generic_fwd);
   m.def("conv_generic_wgrad", &conv_generic_wgrad);

@@ -61,16 +61,17 @@ __device__ __forceinline__ int xcd_swz_tn2(int bid, int nwg) {
 
 struct TnGeom {
   int H, W, Hout, Wout, stride;
+  int kh, kw, pad;  // filter geometry (3,3,1 for the bottleneck 3x3 convs)
 };
 
-// conv3x3 forward gather: output row m=(n,ho,wo) + tap -> X row or -1 (pad)
-__device__ __forceinline__ long tn2_gather3x3(long m, int kh, int kw,
-                                              const TnGeom& g) {
+// generic conv forward gather: output row m=(n,ho,wo) + tap -> X row or -1
+__device__ __forceinline__ long tn2_gather_conv(long m, int tap,
+                                                const TnGeom& g) {
   long t = m;
   const int wo = (int)(t % g.Wout); t /= g.Wout;
   const int ho = (int)(t % g.Hout); t /= g.Hout;
-  const int h = ho * g.stride - 1 + kh;
-  const int w = wo * g.stride - 1 + kw;
+  const int h = ho * g.stride - g.pad + tap / g.kw;
+  const int w = wo * g.stride - g.pad + tap % g.kw;
   if (h < 0 || h >= g.H || w < 0 || w >= g.W) return -1;
   return (t * g.H + h) * g.W + w;
 }
@@ -136,7 +137,7 @@ __device__ __forceinline__ void stage_tn2(
         src = g + tn2_gather_stride(m, geo) * (long)ld + col;
       } else {
         const int tap = col / cin;  // unit never spans taps (cin % 8 == 0)
-        const long row = tn2_gather3x3(m, tap / 3, tap % 3, geo);
+        const long row = tn2_gather_conv(m, tap, geo);
         if (row >= 0) src = g + row * (long)ld + (col - tap * cin);
       }
     }
@@ -276,19 +277,25 @@ tn2_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
       }
 }
 
-// out[i] = sum_p parts[p][i] — fixed-order, vectorized collapse
+// out[i] = sum over p in {slot, slot+step, ...} of parts[p][i] — fixed-order,
+// vectorized.  Used two-stage for large msplit (a single-stage collapse of
+// e.g. 170 partials x 36 K elems leaves only ~36 blocks = latency-bound;
+// stage 1 folds to TN2_FOLD rows in parallel, stage 2 sums those).
+constexpr int TN2_FOLD = 16;
+
 __global__ void __launch_bounds__(AMD_TPB)
 tn2_collapse_kernel(const float* __restrict__ parts, float* __restrict__ out,
-                    int nparts, long NK) {
+                    int nparts, int step, long NK) {
   const long nv = NK / 4;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
-       i += (long)gridDim.x * blockDim.x) {
-    f32x4 s = ((const f32x4*)parts)[i];
-    for (int p = 1; p < nparts; ++p) {
+  const int slot = blockIdx.x % step;
+  for (long i = (long)(blockIdx.x / step) * blockDim.x + threadIdx.x; i < nv;
+       i += (long)(gridDim.x / step) * blockDim.x) {
+    f32x4 s = ((const f32x4*)(parts + (long)slot * NK))[i];
+    for (int p = slot + step; p < nparts; p += step) {
       f32x4 v = ((const f32x4*)(parts + (long)p * NK))[i];
       s.x += v.x; s.y += v.y; s.z += v.z; s.w += v.w;
     }
-    ((f32x4*)out)[i] = s;
+    ((f32x4*)(out + (long)slot * NK))[i] = s;
   }
 }
 
@@ -345,9 +352,23 @@ void tn2_launch(const at::Tensor& dY, const at::Tensor& X, at::Tensor& out,
   if (msplit > 1) {
     const long NK = (long)N * K9;
     TORCH_CHECK(NK % 4 == 0);
-    tn2_collapse_kernel<<<amd_grid(NK / 4), AMD_TPB, 0, stream>>>(
-        (const float*)parts.const_data_ptr(), out.data_ptr<float>(), msplit, NK);
-    CHECK_CUDA_OK();
+    const int nkgrid = amd_grid(NK / 4);
+    if (msplit > 2 * TN2_FOLD) {
+      auto folded = at::empty({TN2_FOLD, NK}, out.options());
+      tn2_collapse_kernel<<<nkgrid * TN2_FOLD, AMD_TPB, 0, stream>>>(
+          (const float*)parts.const_data_ptr(), folded.data_ptr<float>(),
+          msplit, TN2_FOLD, NK);
+      CHECK_CUDA_OK();
+      tn2_collapse_kernel<<<nkgrid, AMD_TPB, 0, stream>>>(
+          (const float*)folded.const_data_ptr(), out.data_ptr<float>(),
+          TN2_FOLD, 1, NK);
+      CHECK_CUDA_OK();
+    } else {
+      tn2_collapse_kernel<<<nkgrid, AMD_TPB, 0, stream>>>(
+          (const float*)parts.const_data_ptr(), out.data_ptr<float>(), msplit,
+          1, NK);
+      CHECK_CUDA_OK();
+    }
   }
 }
 
@@ -358,7 +379,8 @@ void tn2_launch(const at::Tensor& dY, const at::Tensor& X, at::Tensor& out,
 // Returns fp32 [N, taps*Cin]; bitwise deterministic (fixed split + ordered
 // collapse).
 at::Tensor tn2_wgrad(at::Tensor dY, at::Tensor X, long taps, long Nn, long H,
-                     long W, long stride, long gmode) {
+                     long W, long stride, long gmode, long kh, long kw,
+                     long pad) {
   TORCH_CHECK(dY.is_cuda() && dY.scalar_type() == at::kBFloat16 &&
               X.scalar_type() == at::kBFloat16);
   auto Yc = dY.contiguous();
@@ -368,15 +390,17 @@ at::Tensor tn2_wgrad(at::Tensor dY, at::Tensor X, long taps, long Nn, long H,
   const int Cin = (int)Xc.size(1);
   const int K9 = (int)(taps * Cin);
   TORCH_CHECK(Cin % 8 == 0 && N % 8 == 0);
-  TnGeom geo{0, 0, 0, 0, 1};
+  TnGeom geo{0, 0, 0, 0, 1, 3, 3, 1};
   if (gmode == 1) {
     long Hout = (H + stride - 1) / stride, Wout = (W + stride - 1) / stride;
     TORCH_CHECK(M == Nn * Hout * Wout);
-    geo = TnGeom{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
+    geo = TnGeom{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride, 1, 1, 0};
   } else if (gmode == 2) {
-    long Hout = (H + 2 - 3) / stride + 1, Wout = (W + 2 - 3) / stride + 1;
-    TORCH_CHECK(M == Nn * Hout * Wout && taps == 9);
-    geo = TnGeom{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
+    long Hout = (H + 2 * pad - kh) / stride + 1;
+    long Wout = (W + 2 * pad - kw) / stride + 1;
+    TORCH_CHECK(M == Nn * Hout * Wout && taps == kh * kw);
+    geo = TnGeom{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride,
+                 (int)kh, (int)kw, (int)pad};
   } else {
     TORCH_CHECK(M == Xc.size(0));
   }

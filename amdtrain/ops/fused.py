@@ -32,25 +32,28 @@ class _BNFunction(torch.autograd.Function):
                 momentum, eps, relu, addend, parts):
         e = require_ext()
         if parts is not None:
-            y, mean, invstd = e.batch_norm_fwd_train_from_parts(
+            y, mean, invstd, scale, shift = e.batch_norm_fwd_train_from_parts(
                 x, parts, weight, bias, running_mean, running_var,
                 float(momentum), float(eps), bool(relu), addend)
         else:
-            y, mean, invstd = e.batch_norm_fwd_train(
+            y, mean, invstd, scale, shift = e.batch_norm_fwd_train(
                 x, weight, bias, running_mean, running_var,
                 float(momentum), float(eps), bool(relu), addend)
-        ctx.save_for_backward(x, y, weight, mean, invstd)
+        # scale/shift ([C] fp32) let the bwd recompute the relu mask as
+        # scale*x+shift>0 for addend-free BNs — no y read in either bwd pass
+        ctx.save_for_backward(x, y, weight, mean, invstd, scale, shift)
         ctx.relu = bool(relu)
         ctx.has_addend = addend is not None
         return y
 
     @staticmethod
     def backward(ctx, grad_out):
-        x, y, weight, mean, invstd = ctx.saved_tensors
+        x, y, weight, mean, invstd, scale, shift = ctx.saved_tensors
         e = require_ext()
         grad_x, grad_w, grad_b, ghat = e.batch_norm_bwd(
             x, grad_out.contiguous(memory_format=torch.channels_last),
-            y, weight, mean, invstd, ctx.relu, ctx.has_addend)
+            y, weight, mean, invstd, ctx.relu, ctx.has_addend,
+            scale, shift)
         grad_addend = ghat if ctx.has_addend else None
         return (grad_x, grad_w, grad_b, None, None, None, None, None,
                 grad_addend, None)
@@ -77,7 +80,8 @@ def batch_norm(x: torch.Tensor, bn, relu: bool = False,
     with a residual add and/or ReLU."""
     if bn.training and bn.track_running_stats and bn.num_batches_tracked is not None:
         bn.num_batches_tracked.add_(1)
-    if use_ext_for(x):
+    C = x.size(1)
+    if use_ext_for(x) and (C & (C - 1)) == 0:  # kernels take pow2 C only
         parts = getattr(x, "_amdtrain_bn_stats", None) if bn.training else None
         xc = x.contiguous(memory_format=torch.channels_last)
         ac = None if addend is None else addend.contiguous(memory_format=torch.channels_last)
