@@ -29,7 +29,15 @@ import torch.nn as nn
 from ..ops import fused as OF
 
 
-from ..ops.conv import AmdConv2d  # noqa: F401 (stem + helpers)
+import os
+
+from ..ops.conv import AmdConv2d, ResidualGradTap  # noqa: F401
+
+
+def _residual_fuse_enabled() -> bool:
+    """Fuse the identity-shortcut gradient into conv1's dgrad epilogue
+    (AMDTRAIN_RESFUSE=0 restores plain autograd accumulation)."""
+    return os.environ.get("AMDTRAIN_RESFUSE", "1") == "1"
 
 
 def conv3x3(in_planes: int, out_planes: int, stride: int = 1,
@@ -104,11 +112,21 @@ class Bottleneck(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x
+        cell = None
+        if (self.downsample is None and self.training and x.is_cuda
+                and torch.is_grad_enabled() and _residual_fuse_enabled()):
+            # identity-shortcut blocks: route the shortcut gradient into
+            # conv1's dgrad epilogue (ResidualGradTap) instead of an eager
+            # add at x's AccumulateGrad
+            cell = {}
+            self.conv1._amdtrain_grad_cell = cell
         out = self.bn1.forward_relu(self.conv1(x))
         out = self.bn2.forward_relu(self.conv2(out))
         out = self.conv3(out)
         if self.downsample is not None:
             identity = self.downsample(x)
+        elif cell is not None:
+            identity = ResidualGradTap.apply(x, cell)
         return OF.bn_add_relu(out, self.bn3, identity)
 
 

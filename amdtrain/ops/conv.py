@@ -38,14 +38,42 @@ def _wgrad2_enabled() -> bool:
     return os.environ.get("AMDTRAIN_WGRAD2", "1") == "1"
 
 
+class ResidualGradTap(torch.autograd.Function):
+    """Reroutes the identity-shortcut gradient of a residual block into the
+    block's conv1 dgrad epilogue (one fused read) instead of an eager
+    2-read+1-write tensor add at the block input's AccumulateGrad.
+
+    The tap wraps the ADDEND input of bn_add_relu.  Its backward runs
+    strictly before conv1's backward (conv1 is upstream of bn3 in the
+    block, so reverse-topological order guarantees it).  Safety latch: the
+    gradient is stashed ONLY when conv1's forward actually took the fused
+    path and "armed" the shared cell — any fallback keeps plain autograd
+    accumulation, so correctness never depends on the dispatch outcome.
+    """
+
+    @staticmethod
+    def forward(ctx, z: torch.Tensor, cell: dict):
+        ctx.cell = cell
+        return z.view_as(z)
+
+    @staticmethod
+    def backward(ctx, grad: torch.Tensor):
+        if ctx.cell.get("armed"):
+            ctx.cell["g"] = grad
+            return None, None
+        return grad, None
+
+
 class _Conv1x1(torch.autograd.Function):
     """1x1 conv; forward also emits per-block BN-statistics partials
     (non-differentiable 2nd output) for the fused conv->BN pipeline when
-    stride == 1 and fusion is enabled."""
+    stride == 1 and fusion is enabled.  An optional grad_cell (see
+    ResidualGradTap) fuses the residual-branch gradient into the dgrad."""
 
     @staticmethod
     @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
-    def forward(ctx, x: torch.Tensor, weight: torch.Tensor, stride: int):
+    def forward(ctx, x: torch.Tensor, weight: torch.Tensor, stride: int,
+                grad_cell=None):
         e = require_ext()
         n, cin, h, w = x.shape
         cout = weight.shape[0]
@@ -68,6 +96,9 @@ class _Conv1x1(torch.autograd.Function):
             ho, wo = h, w
         ctx.save_for_backward(x2d, w2d)
         ctx.meta = (n, cin, h, w, stride, ho, wo, cout)
+        ctx.grad_cell = grad_cell
+        if grad_cell is not None and stride == 1:
+            grad_cell["armed"] = True
         y = y2d.view(n, ho, wo, cout).permute(0, 3, 1, 2)
         if stats is None:
             stats = torch.empty(0, device=x.device)
@@ -82,9 +113,18 @@ class _Conv1x1(torch.autograd.Function):
         n, cin, h, w, stride, ho, wo, cout = ctx.meta
         gy = grad_y.contiguous(memory_format=torch.channels_last)
         gy2d = _rows(gy).to(torch.bfloat16)
+        # residual-shortcut gradient stashed by ResidualGradTap (fused into
+        # the dgrad epilogue below; stride==1 blocks only)
+        radd = None
+        cell = getattr(ctx, "grad_cell", None)
+        if cell is not None:
+            g = cell.pop("g", None)
+            if g is not None:
+                radd = _rows(g.contiguous(
+                    memory_format=torch.channels_last)).to(torch.bfloat16)
         # dgrad: dX = dY x W  (BT form with pre-transposed weight)
         wT = e.transpose_2d(w2d)                      # [Cin, Cout]
-        dx2d = e.gemm_bt(gy2d, wT, False)             # [R_sub, Cin] bf16
+        dx2d = e.gemm_bt(gy2d, wT, False, radd)       # [R_sub, Cin] bf16
         if stride > 1:
             # fused zero+scatter (one write pass; stride-2 only in ResNet)
             dx = e.scatter_rows_x2(dx2d, n, h, w, ho, wo)
@@ -95,12 +135,12 @@ class _Conv1x1(torch.autograd.Function):
             dx = dx2d.view(n, ho, wo, cin).permute(0, 3, 1, 2)
             dw = e.tn2_wgrad(gy2d, x2d) if _wgrad2_enabled() \
                 else e.gemm_tn(gy2d, x2d, 0)
-        return dx, dw.reshape(cout, cin, 1, 1), None
+        return dx, dw.reshape(cout, cin, 1, 1), None, None
 
 
 def conv1x1_mfma(x: torch.Tensor, weight: torch.Tensor,
-                 stride: int = 1) -> torch.Tensor:
-    y, stats = _Conv1x1.apply(x, weight, stride)
+                 stride: int = 1, grad_cell=None) -> torch.Tensor:
+    y, stats = _Conv1x1.apply(x, weight, stride, grad_cell)
     if stats.numel():
         y._amdtrain_bn_stats = stats
     return y
@@ -249,7 +289,8 @@ class AmdConv2d(nn.Conv2d):
             if (ch_ok and self.kernel_size == (1, 1)
                     and self.padding == (0, 0)
                     and _conv1x1_env_default() == "custom"):
-                return conv1x1_mfma(x, self.weight, self.stride[0])
+                cell = self.__dict__.pop("_amdtrain_grad_cell", None)
+                return conv1x1_mfma(x, self.weight, self.stride[0], cell)
             if (ch_ok and self.kernel_size == (3, 3)
                     and self.padding == (1, 1)
                     and self.stride[0] in (1, 2)

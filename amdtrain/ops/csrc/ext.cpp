@@ -48,7 +48,8 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
                                        std::optional<at::Tensor> shift);
 
 // gemm.hip
-at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out);
+at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out,
+                   std::optional<at::Tensor> addend);
 at::Tensor gemm_tn(at::Tensor dY, at::Tensor X, long msplit);
 at::Tensor gemm_bt_strided(at::Tensor A, at::Tensor B, long Nn, long H,
                            long W, long stride);
@@ -122,7 +123,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("relu"), py::arg("need_ghat"),
         py::arg("scale") = std::nullopt, py::arg("shift") = std::nullopt);
   m.def("gemm_bt", &gemm_bt, py::arg("A"), py::arg("B"),
-        py::arg("f32_out") = false);
+        py::arg("f32_out") = false, py::arg("addend") = std::nullopt);
   m.def("gemm_tn", &gemm_tn, py::arg("dY"), py::arg("X"),
         py::arg("msplit") = 0);
   m.def("gemm_bt_strided", &gemm_bt_strided);

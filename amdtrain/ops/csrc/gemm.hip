@@ -161,7 +161,8 @@ template <bool F32OUT, bool STRIDED>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                void* __restrict__ C, long M, long N, long K, int nbm,
-               int nbn, StrideMap sm, float* __restrict__ stats) {
+               int nbn, StrideMap sm, float* __restrict__ stats,
+               const bf16* __restrict__ gadd) {
   __shared__ bf16 As[BM * BK];
   __shared__ bf16 Bs[BN * BK];
 
@@ -219,7 +220,9 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
             a[i], b[j], acc[i][j], 0, 0, 0);
   }
 
-  // epilogue: C[m0+wm+i*16+fq*4+r][n0+wn+j*16+fr]
+  // epilogue: C[m0+wm+i*16+fq*4+r][n0+wn+j*16+fr]; the optional gadd
+  // addend fuses the residual-branch gradient accumulation into the
+  // conv1 dgrad (one extra read replaces an eager 2-read+1-write add)
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
 #pragma unroll
@@ -229,10 +232,13 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
         long row = m0 + wm + i * 16 + fq * 4 + r;
         long col = n0 + wn + j * 16 + fr;
         if (row < M && col < N) {
+          float v = acc[i][j][r];
+          if (gadd != nullptr)
+            v += __bfloat162float(gadd[row * N + col]);
           if (F32OUT)
-            ((float*)C)[row * N + col] = acc[i][j][r];
+            ((float*)C)[row * N + col] = v;
           else
-            ((bf16*)C)[row * N + col] = __float2bfloat16(acc[i][j][r]);
+            ((bf16*)C)[row * N + col] = __float2bfloat16(v);
         }
       }
     }
@@ -406,7 +412,8 @@ transpose_2d_kernel(const bf16* __restrict__ in, bf16* __restrict__ out,
 
 }  // namespace
 
-at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out) {
+at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out,
+                   std::optional<at::Tensor> addend) {
   TORCH_CHECK(A.is_cuda() && A.dim() == 2 && B.dim() == 2);
   TORCH_CHECK(A.scalar_type() == at::kBFloat16 &&
               B.scalar_type() == at::kBFloat16,
@@ -421,14 +428,22 @@ at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out) {
   int nbm = (int)((M + BM - 1) / BM), nbn = (int)((N + BN - 1) / BN);
   auto stream = at::cuda::getCurrentCUDAStream();
   StrideMap sm{0, 0, 0, 0, 1};
+  const bf16* gadd = nullptr;
+  at::Tensor ac;
+  if (addend) {
+    ac = addend->contiguous();
+    TORCH_CHECK(ac.scalar_type() == at::kBFloat16 &&
+                ac.numel() == M * N, "addend must be bf16 [M,N]");
+    gadd = (const bf16*)ac.const_data_ptr();
+  }
   if (f32_out)
     gemm_bt_kernel<true, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-        C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr);
+        C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr, gadd);
   else
     gemm_bt_kernel<false, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-        C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr);
+        C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr, gadd);
   CHECK_CUDA_OK();
   return C;
 }
@@ -448,7 +463,7 @@ std::vector<at::Tensor> gemm_bt_stats(at::Tensor A, at::Tensor B) {
   auto stream = at::cuda::getCurrentCUDAStream();
   gemm_bt_kernel<false, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
       (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-      C.data_ptr(), M, N, K, nbm, nbn, sm, stats.data_ptr<float>());
+      C.data_ptr(), M, N, K, nbm, nbn, sm, stats.data_ptr<float>(), nullptr);
   CHECK_CUDA_OK();
   return {C, stats};
 }
@@ -472,7 +487,7 @@ at::Tensor gemm_bt_strided(at::Tensor A, at::Tensor B, long Nn, long H,
   auto stream = at::cuda::getCurrentCUDAStream();
   gemm_bt_kernel<false, true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
       (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-      C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr);
+      C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr, nullptr);
   CHECK_CUDA_OK();
   return C;
 }
