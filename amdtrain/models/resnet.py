@@ -32,6 +32,7 @@ from ..ops import fused as OF
 import os
 
 from ..ops.conv import AmdConv2d, ResidualGradTap  # noqa: F401
+from ..ops.linear import AmdLinear
 
 
 def _residual_fuse_enabled() -> bool:
@@ -149,7 +150,7 @@ class ResNet(nn.Module):
         self.layer3 = self._make_layer(block, 256, layers[2], stride=2)
         self.layer4 = self._make_layer(block, 512, layers[3], stride=2)
         self.avgpool = nn.AdaptiveAvgPool2d((1, 1))
-        self.fc = nn.Linear(512 * block.expansion, num_classes)
+        self.fc = AmdLinear(512 * block.expansion, num_classes)
 
         for m in self.modules():
             if isinstance(m, nn.Conv2d):

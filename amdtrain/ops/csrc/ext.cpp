@@ -49,7 +49,8 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
 
 // gemm.hip
 at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out,
-                   std::optional<at::Tensor> addend);
+                   std::optional<at::Tensor> addend,
+                   std::optional<at::Tensor> bias);
 at::Tensor gemm_tn(at::Tensor dY, at::Tensor X, long msplit);
 at::Tensor gemm_bt_strided(at::Tensor A, at::Tensor B, long Nn, long H,
                            long W, long stride);
@@ -123,7 +124,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("relu"), py::arg("need_ghat"),
         py::arg("scale") = std::nullopt, py::arg("shift") = std::nullopt);
   m.def("gemm_bt", &gemm_bt, py::arg("A"), py::arg("B"),
-        py::arg("f32_out") = false, py::arg("addend") = std::nullopt);
+        py::arg("f32_out") = false, py::arg("addend") = std::nullopt,
+        py::arg("bias") = std::nullopt);
   m.def("gemm_tn", &gemm_tn, py::arg("dY"), py::arg("X"),
         py::arg("msplit") = 0);
   m.def("gemm_bt_strided", &gemm_bt_strided);

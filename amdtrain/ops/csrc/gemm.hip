@@ -162,7 +162,8 @@ __global__ void __launch_bounds__(GEMM_TPB, 2)
 gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                void* __restrict__ C, long M, long N, long K, int nbm,
                int nbn, StrideMap sm, float* __restrict__ stats,
-               const bf16* __restrict__ gadd) {
+               const bf16* __restrict__ gadd,
+               const float* __restrict__ bias) {
   __shared__ bf16 As[BM * BK];
   __shared__ bf16 Bs[BN * BK];
 
@@ -235,6 +236,7 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
           float v = acc[i][j][r];
           if (gadd != nullptr)
             v += __bfloat162float(gadd[row * N + col]);
+          if (bias != nullptr) v += bias[col];
           if (F32OUT)
             ((float*)C)[row * N + col] = v;
           else
@@ -413,7 +415,8 @@ transpose_2d_kernel(const bf16* __restrict__ in, bf16* __restrict__ out,
 }  // namespace
 
 at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out,
-                   std::optional<at::Tensor> addend) {
+                   std::optional<at::Tensor> addend,
+                   std::optional<at::Tensor> bias) {
   TORCH_CHECK(A.is_cuda() && A.dim() == 2 && B.dim() == 2);
   TORCH_CHECK(A.scalar_type() == at::kBFloat16 &&
               B.scalar_type() == at::kBFloat16,
@@ -429,21 +432,27 @@ at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out,
   auto stream = at::cuda::getCurrentCUDAStream();
   StrideMap sm{0, 0, 0, 0, 1};
   const bf16* gadd = nullptr;
-  at::Tensor ac;
+  at::Tensor ac, bc2;
   if (addend) {
     ac = addend->contiguous();
     TORCH_CHECK(ac.scalar_type() == at::kBFloat16 &&
                 ac.numel() == M * N, "addend must be bf16 [M,N]");
     gadd = (const bf16*)ac.const_data_ptr();
   }
+  const float* biasp = nullptr;
+  if (bias) {
+    bc2 = bias->to(at::kFloat).contiguous();
+    TORCH_CHECK(bc2.numel() == N, "bias must be [N]");
+    biasp = (const float*)bc2.const_data_ptr();
+  }
   if (f32_out)
     gemm_bt_kernel<true, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-        C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr, gadd);
+        C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr, gadd, biasp);
   else
     gemm_bt_kernel<false, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-        C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr, gadd);
+        C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr, gadd, biasp);
   CHECK_CUDA_OK();
   return C;
 }
@@ -463,7 +472,8 @@ std::vector<at::Tensor> gemm_bt_stats(at::Tensor A, at::Tensor B) {
   auto stream = at::cuda::getCurrentCUDAStream();
   gemm_bt_kernel<false, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
       (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-      C.data_ptr(), M, N, K, nbm, nbn, sm, stats.data_ptr<float>(), nullptr);
+      C.data_ptr(), M, N, K, nbm, nbn, sm, stats.data_ptr<float>(), nullptr,
+      nullptr);
   CHECK_CUDA_OK();
   return {C, stats};
 }
@@ -487,7 +497,7 @@ at::Tensor gemm_bt_strided(at::Tensor A, at::Tensor B, long Nn, long H,
   auto stream = at::cuda::getCurrentCUDAStream();
   gemm_bt_kernel<false, true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
       (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-      C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr, nullptr);
+      C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr, nullptr, nullptr);
   CHECK_CUDA_OK();
   return C;
 }

@@ -155,3 +155,31 @@ def test_conv3x3_8p_matches_old():
     y_old = e.conv3x3_fwd(x2d, n, hw, hw, 1, w2d)
     y_8p = e.conv3x3_8p(x2d, n, hw, hw, 1, w2d, False)
     assert torch.allclose(y_old.float(), y_8p.float(), atol=1e-2)
+
+
+def test_amd_linear_parity():
+    """FC classifier (fwd bias epilogue + dgrad pad + tn2 wgrad) vs fp32
+    F.linear — retires the last library GEMM on the hot path."""
+    import torch.nn.functional as F
+    from amdtrain.ops.linear import AmdLinear
+
+    torch.manual_seed(0)
+    lin = AmdLinear(2048, 1000).cuda()
+    x = torch.randn(256, 2048, device="cuda")
+    # fp32 reference
+    xr = x.clone().requires_grad_(True)
+    wr = lin.weight.detach().clone().requires_grad_(True)
+    br = lin.bias.detach().clone().requires_grad_(True)
+    yr = F.linear(xr, wr, br)
+    gy = torch.randn_like(yr)
+    yr.backward(gy)
+    # custom path under autocast
+    xc = x.clone().requires_grad_(True)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        y = lin(xc)
+    assert y.dtype == torch.bfloat16
+    y.backward(gy.bfloat16())
+    assert torch.allclose(y.float(), yr, atol=2.0, rtol=0.05)
+    assert torch.allclose(xc.grad.float(), xr.grad, atol=2.0, rtol=0.05)
+    assert torch.allclose(lin.weight.grad, wr.grad, atol=2.0, rtol=0.05)
+    assert torch.allclose(lin.bias.grad, br.grad, atol=1.0, rtol=0.05)
