@@ -31,7 +31,7 @@ from ..ops import fused as OF
 
 import os
 
-from ..ops.conv import AmdConv2d, ResidualGradTap  # noqa: F401
+from ..ops.conv import AmdConv2d, GradCell, ResidualGradTap  # noqa: F401
 from ..ops.linear import AmdLinear
 
 
@@ -119,7 +119,7 @@ class Bottleneck(nn.Module):
             # identity-shortcut blocks: route the shortcut gradient into
             # conv1's dgrad epilogue (ResidualGradTap) instead of an eager
             # add at x's AccumulateGrad
-            cell = {}
+            cell = GradCell()
             self.conv1._amdtrain_grad_cell = cell
         out = self.bn1.forward_relu(self.conv1(x))
         out = self.bn2.forward_relu(self.conv2(out))

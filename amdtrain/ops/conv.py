@@ -38,6 +38,18 @@ def _wgrad2_enabled() -> bool:
     return os.environ.get("AMDTRAIN_WGRAD2", "1") == "1"
 
 
+class GradCell:
+    """Shared mailbox between ResidualGradTap and _Conv1x1.  A plain class
+    (NOT a dict): torch.amp.custom_fwd(cast_inputs=...) deep-copies dict
+    arguments while casting, which would silently disconnect the tap from
+    the conv.  Class instances pass through the cast untouched."""
+    __slots__ = ("armed", "g")
+
+    def __init__(self):
+        self.armed = False
+        self.g = None
+
+
 class ResidualGradTap(torch.autograd.Function):
     """Reroutes the identity-shortcut gradient of a residual block into the
     block's conv1 dgrad epilogue (one fused read) instead of an eager
@@ -52,14 +64,14 @@ class ResidualGradTap(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, z: torch.Tensor, cell: dict):
+    def forward(ctx, z: torch.Tensor, cell: "GradCell"):
         ctx.cell = cell
         return z.view_as(z)
 
     @staticmethod
     def backward(ctx, grad: torch.Tensor):
-        if ctx.cell.get("armed"):
-            ctx.cell["g"] = grad
+        if ctx.cell.armed:
+            ctx.cell.g = grad
             return None, None
         return grad, None
 
@@ -98,7 +110,7 @@ class _Conv1x1(torch.autograd.Function):
         ctx.meta = (n, cin, h, w, stride, ho, wo, cout)
         ctx.grad_cell = grad_cell
         if grad_cell is not None and stride == 1:
-            grad_cell["armed"] = True
+            grad_cell.armed = True
         y = y2d.view(n, ho, wo, cout).permute(0, 3, 1, 2)
         if stats is None:
             stats = torch.empty(0, device=x.device)
@@ -117,11 +129,10 @@ class _Conv1x1(torch.autograd.Function):
         # the dgrad epilogue below; stride==1 blocks only)
         radd = None
         cell = getattr(ctx, "grad_cell", None)
-        if cell is not None:
-            g = cell.pop("g", None)
-            if g is not None:
-                radd = _rows(g.contiguous(
-                    memory_format=torch.channels_last)).to(torch.bfloat16)
+        if cell is not None and cell.g is not None:
+            radd = _rows(cell.g.contiguous(
+                memory_format=torch.channels_last)).to(torch.bfloat16)
+            cell.g = None
         # dgrad: dX = dY x W  (BT form with pre-transposed weight)
         wT = e.transpose_2d(w2d)                      # [Cin, Cout]
         dx2d = e.gemm_bt(gy2d, wT, False, radd)       # [R_sub, Cin] bf16

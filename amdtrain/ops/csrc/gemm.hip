@@ -250,6 +250,92 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                       fr, fq, nbm, As);
 }
 
+// ---- split-K BT GEMM for tiny-M shapes (the FC classifier) ---------------
+// At [B,2048]x[2048,1000] the plain kernel has only ceil(B/128)*8 blocks —
+// a fraction of the 256-CU chip — while K=2048 runs 64 serial K-steps.
+// Split the K range across ksplit block-groups into fp32 partials, then
+// collapse (+bias/addend, cast) in one small pass.
+__global__ void __launch_bounds__(GEMM_TPB, 2)
+gemm_bt_ks_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
+                  float* __restrict__ parts, long M, long N, long K, int nbm,
+                  int nbn, int ksplit) {
+  __shared__ bf16 As[BM * BK];
+  __shared__ bf16 Bs[BN * BK];
+
+  const int tiles = nbm * nbn;
+  const int bid = xcd_swizzle(blockIdx.x, tiles * ksplit);
+  const int tile = bid % tiles;
+  const int kp = bid / tiles;
+  const int bm = tile / nbn, bn = tile % nbn;
+  const long m0 = (long)bm * BM, n0 = (long)bn * BN;
+
+  const long ksteps = K / BK;
+  const long kper = (ksteps + ksplit - 1) / ksplit;
+  const long kt0 = kp * kper;
+  const long kt1 = min(kt0 + kper, ksteps);
+
+  const int t = threadIdx.x;
+  const int wave = t / AMD_WAVE;
+  const int lane = t % AMD_WAVE;
+  const int wm = (wave >> 1) * 64;
+  const int wn = (wave & 1) * 64;
+  const int fr = lane & 15;
+  const int fq = lane >> 4;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (long kt = kt0; kt < kt1; ++kt) {
+    __syncthreads();
+    stage_tile_128x32(A, K, m0, M, kt * BK, As);
+    stage_tile_128x32(B, K, n0, N, kt * BK, Bs);
+    __syncthreads();
+    bf16x8 a[4], b[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      a[i] = *(const bf16x8*)&As[(wm + i * 16 + fr) * BK + fq * 8];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      b[j] = *(const bf16x8*)&Bs[(wn + j * 16 + fr) * BK + fq * 8];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a[i], b[j], acc[i][j], 0, 0, 0);
+  }
+
+  float* out = parts + (long)kp * M * N;
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = m0 + wm + i * 16 + fq * 4 + r;
+        long col = n0 + wn + j * 16 + fr;
+        if (row < M && col < N) out[row * N + col] = acc[i][j][r];
+      }
+}
+
+__global__ void __launch_bounds__(AMD_TPB)
+gemm_ks_collapse_kernel(const float* __restrict__ parts,
+                        bf16* __restrict__ out, int ksplit, long MN, long N,
+                        const float* __restrict__ bias,
+                        const bf16* __restrict__ gadd) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < MN;
+       i += (long)gridDim.x * blockDim.x) {
+    float v = parts[i];
+    for (int p = 1; p < ksplit; ++p) v += parts[(long)p * MN + i];
+    if (bias != nullptr) v += bias[i % N];
+    if (gadd != nullptr) v += __bfloat162float(gadd[i]);
+    out[i] = __float2bfloat16(v);
+  }
+}
+
 // ---- TN GEMM for wgrad: dW[N,K] += sum_m dY[m,n] * X[m,k] ----------------
 // Output tiles 128n x 128k, M chunked by 32 and split across blocks
 // (split-M) with fp32 atomic accumulation into dW.
@@ -444,6 +530,23 @@ at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out,
     bc2 = bias->to(at::kFloat).contiguous();
     TORCH_CHECK(bc2.numel() == N, "bias must be [N]");
     biasp = (const float*)bc2.const_data_ptr();
+  }
+  const long tiles = (long)nbm * nbn;
+  const long ksteps = K / BK;
+  if (!f32_out && tiles <= 128 && ksteps >= 2 && 512 / tiles >= 2) {
+    // tiny-M path (FC classifier fwd/dgrad): split K to fill the chip
+    const int ksplit = (int)std::min<long>(ksteps, 512 / tiles);
+    auto parts = at::empty({ksplit, M * N},
+                           Ac.options().dtype(at::kFloat));
+    gemm_bt_ks_kernel<<<(int)(tiles * ksplit), GEMM_TPB, 0, stream>>>(
+        (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
+        parts.data_ptr<float>(), M, N, K, nbm, nbn, ksplit);
+    CHECK_CUDA_OK();
+    gemm_ks_collapse_kernel<<<amd_grid(M * N), AMD_TPB, 0, stream>>>(
+        (const float*)parts.const_data_ptr(), (bf16*)C.data_ptr(), ksplit,
+        M * N, N, biasp, gadd);
+    CHECK_CUDA_OK();
+    return C;
   }
   if (f32_out)
     gemm_bt_kernel<true, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
