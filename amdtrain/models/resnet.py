@@ -84,10 +84,17 @@ class BasicBlock(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x
+        cell = None
+        if (self.downsample is None and self.training and x.is_cuda
+                and torch.is_grad_enabled() and _residual_fuse_enabled()):
+            cell = getattr(x, "_amdtrain_next_cell", None)
         out = self.bn1.forward_relu(self.conv1(x))
         out = self.conv2(out)
         if self.downsample is not None:
             identity = self.downsample(x)
+        elif cell is not None:
+            cell.armed = True
+            identity = ResidualGradTap.apply(x, cell)
         return OF.bn_add_relu(out, self.bn2, identity)
 
 
@@ -116,17 +123,17 @@ class Bottleneck(nn.Module):
         cell = None
         if (self.downsample is None and self.training and x.is_cuda
                 and torch.is_grad_enabled() and _residual_fuse_enabled()):
-            # identity-shortcut blocks: route the shortcut gradient into
-            # conv1's dgrad epilogue (ResidualGradTap) instead of an eager
-            # add at x's AccumulateGrad
-            cell = GradCell()
-            self.conv1._amdtrain_grad_cell = cell
+            # identity-shortcut block: reroute the shortcut gradient into
+            # the PRODUCING block-tail BN's backward (mailbox attached to
+            # x by that BN) — kills the eager add at x's AccumulateGrad
+            cell = getattr(x, "_amdtrain_next_cell", None)
         out = self.bn1.forward_relu(self.conv1(x))
         out = self.bn2.forward_relu(self.conv2(out))
         out = self.conv3(out)
         if self.downsample is not None:
             identity = self.downsample(x)
         elif cell is not None:
+            cell.armed = True
             identity = ResidualGradTap.apply(x, cell)
         return OF.bn_add_relu(out, self.bn3, identity)
 

@@ -39,10 +39,10 @@ def _wgrad2_enabled() -> bool:
 
 
 class GradCell:
-    """Shared mailbox between ResidualGradTap and _Conv1x1.  A plain class
-    (NOT a dict): torch.amp.custom_fwd(cast_inputs=...) deep-copies dict
-    arguments while casting, which would silently disconnect the tap from
-    the conv.  Class instances pass through the cast untouched."""
+    """Shared mailbox between ResidualGradTap and the producing block-tail
+    _BNFunction (fused.py).  A plain class (NOT a dict):
+    torch.amp.custom_fwd(cast_inputs=...) deep-copies dict arguments while
+    casting, which would silently disconnect the mailbox."""
     __slots__ = ("armed", "g")
 
     def __init__(self):
@@ -51,16 +51,20 @@ class GradCell:
 
 
 class ResidualGradTap(torch.autograd.Function):
-    """Reroutes the identity-shortcut gradient of a residual block into the
-    block's conv1 dgrad epilogue (one fused read) instead of an eager
-    2-read+1-write tensor add at the block input's AccumulateGrad.
+    """Reroutes a residual block's identity-shortcut gradient into the
+    PRODUCING block-tail BN's backward (a second, linearly-read go operand
+    of its reduce kernel) instead of an eager 2-read+1-write tensor add at
+    the shared tensor's AccumulateGrad.
 
-    The tap wraps the ADDEND input of bn_add_relu.  Its backward runs
-    strictly before conv1's backward (conv1 is upstream of bn3 in the
-    block, so reverse-topological order guarantees it).  Safety latch: the
-    gradient is stashed ONLY when conv1's forward actually took the fused
-    path and "armed" the shared cell — any fallback keeps plain autograd
-    accumulation, so correctness never depends on the dispatch outcome.
+    The tap wraps the ADDEND input of block b's bn_add_relu; the cell is
+    created and attached by block b-1's tail BN, whose backward runs
+    strictly AFTER every node of block b (it is upstream of all of them).
+    A first attempt fused this add into conv1's dgrad epilogue instead:
+    measured SLOWER (the fragment-shaped 2-byte epilogue reads cost ~5x
+    the eager add) — the BN reduce reads the extra operand with the same
+    16 B vectorized stream as its other inputs.  Safety latch: the grad is
+    stashed ONLY when the producing BN armed the cell; any fallback keeps
+    plain autograd accumulation.
     """
 
     @staticmethod
@@ -84,8 +88,7 @@ class _Conv1x1(torch.autograd.Function):
 
     @staticmethod
     @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
-    def forward(ctx, x: torch.Tensor, weight: torch.Tensor, stride: int,
-                grad_cell=None):
+    def forward(ctx, x: torch.Tensor, weight: torch.Tensor, stride: int):
         e = require_ext()
         n, cin, h, w = x.shape
         cout = weight.shape[0]
@@ -108,9 +111,6 @@ class _Conv1x1(torch.autograd.Function):
             ho, wo = h, w
         ctx.save_for_backward(x2d, w2d)
         ctx.meta = (n, cin, h, w, stride, ho, wo, cout)
-        ctx.grad_cell = grad_cell
-        if grad_cell is not None and stride == 1:
-            grad_cell.armed = True
         y = y2d.view(n, ho, wo, cout).permute(0, 3, 1, 2)
         if stats is None:
             stats = torch.empty(0, device=x.device)
@@ -125,17 +125,9 @@ class _Conv1x1(torch.autograd.Function):
         n, cin, h, w, stride, ho, wo, cout = ctx.meta
         gy = grad_y.contiguous(memory_format=torch.channels_last)
         gy2d = _rows(gy).to(torch.bfloat16)
-        # residual-shortcut gradient stashed by ResidualGradTap (fused into
-        # the dgrad epilogue below; stride==1 blocks only)
-        radd = None
-        cell = getattr(ctx, "grad_cell", None)
-        if cell is not None and cell.g is not None:
-            radd = _rows(cell.g.contiguous(
-                memory_format=torch.channels_last)).to(torch.bfloat16)
-            cell.g = None
         # dgrad: dX = dY x W  (BT form with pre-transposed weight)
         wT = e.transpose_2d(w2d)                      # [Cin, Cout]
-        dx2d = e.gemm_bt(gy2d, wT, False, radd)       # [R_sub, Cin] bf16
+        dx2d = e.gemm_bt(gy2d, wT, False)             # [R_sub, Cin] bf16
         if stride > 1:
             # fused zero+scatter (one write pass; stride-2 only in ResNet)
             dx = e.scatter_rows_x2(dx2d, n, h, w, ho, wo)
@@ -146,12 +138,12 @@ class _Conv1x1(torch.autograd.Function):
             dx = dx2d.view(n, ho, wo, cin).permute(0, 3, 1, 2)
             dw = e.tn2_wgrad(gy2d, x2d) if _wgrad2_enabled() \
                 else e.gemm_tn(gy2d, x2d, 0)
-        return dx, dw.reshape(cout, cin, 1, 1), None, None
+        return dx, dw.reshape(cout, cin, 1, 1), None
 
 
 def conv1x1_mfma(x: torch.Tensor, weight: torch.Tensor,
-                 stride: int = 1, grad_cell=None) -> torch.Tensor:
-    y, stats = _Conv1x1.apply(x, weight, stride, grad_cell)
+                 stride: int = 1) -> torch.Tensor:
+    y, stats = _Conv1x1.apply(x, weight, stride)
     if stats.numel():
         y._amdtrain_bn_stats = stats
     return y
@@ -300,8 +292,7 @@ class AmdConv2d(nn.Conv2d):
             if (ch_ok and self.kernel_size == (1, 1)
                     and self.padding == (0, 0)
                     and _conv1x1_env_default() == "custom"):
-                cell = self.__dict__.pop("_amdtrain_grad_cell", None)
-                return conv1x1_mfma(x, self.weight, self.stride[0], cell)
+                return conv1x1_mfma(x, self.weight, self.stride[0])
             if (ch_ok and self.kernel_size == (3, 3)
                     and self.padding == (1, 1)
                     and self.stride[0] in (1, 2)

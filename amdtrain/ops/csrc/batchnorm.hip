@@ -28,10 +28,14 @@ namespace {
 // WG: additionally materialize ghat (consumed by the mask-free bwd apply and
 // as the residual addend gradient) — turns reduce into 3r+1w so apply drops
 // to 2r+1w (guide: every pass here is an HBM-bound R x C stream).
-template <typename T, int VEC, bool BWD, int MASK, bool WG>
+// GO2: go is the SUM of two operands (goB = the rerouted residual
+// shortcut gradient, see ResidualGradTap) — added in fp32 here so the
+// eager CUDAFunctor_add pass disappears; the sum is baked into ghat_out.
+template <typename T, int VEC, bool BWD, int MASK, bool WG, bool GO2>
 __global__ void __launch_bounds__(AMD_TPB)
 bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
-                 const T* __restrict__ y, const float* __restrict__ mean,
+                 const T* __restrict__ goB, const T* __restrict__ y,
+                 const float* __restrict__ mean,
                  const float* __restrict__ invstd,
                  const float* __restrict__ scale,
                  const float* __restrict__ shift, T* __restrict__ ghat_out,
@@ -67,9 +71,10 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
     for (long r = r0 + phase; r < r1; r += rstep) {
       const long base = r * C + c0;
       Pack<T, VEC> xv = *(const Pack<T, VEC>*)(x + base);
-      Pack<T, VEC> gv, yv, gh;
+      Pack<T, VEC> gv, gv2, yv, gh;
       if (BWD) {
         gv = *(const Pack<T, VEC>*)(go + base);
+        if (GO2) gv2 = *(const Pack<T, VEC>*)(goB + base);
         if (MASK == 1) yv = *(const Pack<T, VEC>*)(y + base);
       }
 #pragma unroll
@@ -77,6 +82,7 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
         float xe = to_f32(xv.v[k]);
         if (BWD) {
           float ge = to_f32(gv.v[k]);
+          if (GO2) ge += to_f32(gv2.v[k]);
           if (MASK == 1 && to_f32(yv.v[k]) <= 0.f) ge = 0.f;
           if (MASK == 2 && sck[k] * xe + shk[k] <= 0.f) ge = 0.f;
           sa[k] += ge;
@@ -142,9 +148,10 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
         const int c0 = (t + j * AMD_TPB) * VEC;
         const long base = r * C + c0;
         Pack<T, VEC> xv = *(const Pack<T, VEC>*)(x + base);
-        Pack<T, VEC> gv, yv, gh;
+        Pack<T, VEC> gv, gv2, yv, gh;
         if (BWD) {
           gv = *(const Pack<T, VEC>*)(go + base);
+          if (GO2) gv2 = *(const Pack<T, VEC>*)(goB + base);
           if (MASK == 1) yv = *(const Pack<T, VEC>*)(y + base);
         }
 #pragma unroll
@@ -152,6 +159,7 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
           float xe = to_f32(xv.v[k]);
           if (BWD) {
             float ge = to_f32(gv.v[k]);
+            if (GO2) ge += to_f32(gv2.v[k]);
             if (MASK == 1 && to_f32(yv.v[k]) <= 0.f) ge = 0.f;
             if (MASK == 2 && sck[j][k] * xe + shk[j][k] <= 0.f) ge = 0.f;
             sa[j][k] += ge;
@@ -426,10 +434,10 @@ std::vector<at::Tensor> batch_norm_fwd_train(
   dispatch_vec(x, [&](auto* tp, auto vec) {
     using devT = std::remove_pointer_t<decltype(tp)>;
     constexpr int VEC = decltype(vec)::value;
-    bn_reduce_kernel<devT, VEC, false, 0, false>
+    bn_reduce_kernel<devT, VEC, false, 0, false, false>
         <<<rgrid, AMD_TPB, 0, stream>>>((const devT*)x.const_data_ptr(),
                                         nullptr, nullptr, nullptr, nullptr,
-                                        nullptr, nullptr, nullptr,
+                                        nullptr, nullptr, nullptr, nullptr,
                                         sums.data_ptr<float>(), R, (int)C);
     CHECK_CUDA_OK();
     auto sums2 = at::empty({BN_COLLAPSE, 2 * C}, opts);
@@ -575,7 +583,8 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
                                        at::Tensor mean, at::Tensor invstd,
                                        bool relu, bool need_ghat,
                                        std::optional<at::Tensor> scale,
-                                       std::optional<at::Tensor> shift) {
+                                       std::optional<at::Tensor> shift,
+                                       std::optional<at::Tensor> grad_out2) {
   // Pass structure (every kernel is an HBM-bound R x C stream, so passes
   // are the whole cost):
   //  * no relu:                reduce(x,go) -> apply(x,go -> gx)
@@ -600,9 +609,11 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
   auto Dc = at::empty({C}, opts);
   auto gx = at::empty_like(x);
   const bool affine_mask = relu && !need_ghat && scale && shift;
-  const bool mask_y = relu && !affine_mask;  // reduce masks from y + writes ghat
+  // grad_out2 (rerouted residual gradient) forces the ghat-writing path so
+  // the fp32 sum is materialized once
+  const bool mask_y = relu && (!affine_mask || grad_out2.has_value());
   at::Tensor ghat;
-  if (mask_y) ghat = at::empty_like(grad_out);
+  if (mask_y || grad_out2) ghat = at::empty_like(grad_out);
   auto stream = at::cuda::getCurrentCUDAStream();
 
   dispatch_vec(x, [&](auto* tp, auto vec) {
@@ -610,18 +621,22 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
     constexpr int VEC = decltype(vec)::value;
     const float* scp = affine_mask ? scale->data_ptr<float>() : nullptr;
     const float* shp = affine_mask ? shift->data_ptr<float>() : nullptr;
-#define REDUCE(MASK_, WG_)                                                  \
-  bn_reduce_kernel<devT, VEC, true, MASK_, WG_>                             \
+    const devT* go2p = grad_out2
+        ? (const devT*)grad_out2->const_data_ptr() : nullptr;
+#define REDUCE(MASK_, WG_, GO2_)                                            \
+  bn_reduce_kernel<devT, VEC, true, MASK_, WG_, GO2_>                       \
       <<<rgrid, AMD_TPB, 0, stream>>>(                                      \
           (const devT*)x.const_data_ptr(),                                  \
-          (const devT*)grad_out.const_data_ptr(),                           \
+          (const devT*)grad_out.const_data_ptr(), go2p,                     \
           (const devT*)y.const_data_ptr(), mean.data_ptr<float>(),          \
           invstd.data_ptr<float>(), scp, shp,                               \
           WG_ ? (devT*)ghat.data_ptr() : nullptr,                           \
           sums.data_ptr<float>(), R, (int)C)
-    if (mask_y) REDUCE(1, true);
-    else if (affine_mask) REDUCE(2, false);
-    else REDUCE(0, false);
+    if (mask_y && go2p) REDUCE(1, true, true);
+    else if (mask_y) REDUCE(1, true, false);
+    else if (affine_mask) REDUCE(2, false, false);
+    else if (go2p) REDUCE(0, true, true);
+    else REDUCE(0, false, false);
 #undef REDUCE
     CHECK_CUDA_OK();
     auto sums2 = at::empty({BN_COLLAPSE, 2 * C}, opts);
@@ -642,8 +657,11 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
     long total_vec = R * C / VEC;
     int agrid = amd_grid(total_vec);
     size_t smem = (affine_mask ? 5 : 3) * C * sizeof(float);
-    const devT* go_in = mask_y ? (const devT*)ghat.const_data_ptr()
-                               : (const devT*)grad_out.const_data_ptr();
+    // use the materialized ghat whenever the reduce produced it (mask or
+    // summed grad_out2); otherwise the raw grad_out
+    const devT* go_in = ghat.defined()
+                            ? (const devT*)ghat.const_data_ptr()
+                            : (const devT*)grad_out.const_data_ptr();
 #define BAPPLY(MASK_)                                                       \
   bn_bwd_apply_kernel<devT, VEC, MASK_, false, true>                        \
       <<<agrid, AMD_TPB, smem, stream>>>(                                   \
@@ -655,6 +673,6 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
 #undef BAPPLY
     CHECK_CUDA_OK();
   });
-  if (!mask_y) ghat = grad_out;  // placeholder / identity (no addend)
+  if (!ghat.defined()) ghat = grad_out;  // placeholder (no addend path)
   return {gx, gw, gb, ghat};
 }

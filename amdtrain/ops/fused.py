@@ -18,6 +18,7 @@ import torch
 import torch.nn.functional as F
 
 from ._ext import require_ext, use_ext_for
+from .conv import GradCell
 
 __all__ = ["batch_norm", "bn_add_relu", "max_pool_3x3_s2", "global_avg_pool"]
 
@@ -29,7 +30,7 @@ class _BNFunction(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var,
-                momentum, eps, relu, addend, parts):
+                momentum, eps, relu, addend, parts, res_cell=None):
         e = require_ext()
         if parts is not None:
             y, mean, invstd, scale, shift = e.batch_norm_fwd_train_from_parts(
@@ -44,19 +45,29 @@ class _BNFunction(torch.autograd.Function):
         ctx.save_for_backward(x, y, weight, mean, invstd, scale, shift)
         ctx.relu = bool(relu)
         ctx.has_addend = addend is not None
+        # block-tail BN grad mailbox (see batch_norm below): our backward
+        # folds a rerouted shortcut gradient into the reduce as a second
+        # linear go operand, replacing the eager CUDAFunctor_add at the
+        # output's AccumulateGrad.
+        ctx.res_cell = res_cell
         return y
 
     @staticmethod
     def backward(ctx, grad_out):
         x, y, weight, mean, invstd, scale, shift = ctx.saved_tensors
         e = require_ext()
+        go2 = None
+        cell = getattr(ctx, "res_cell", None)
+        if cell is not None and cell.g is not None:
+            go2 = cell.g.contiguous(memory_format=torch.channels_last)
+            cell.g = None
         grad_x, grad_w, grad_b, ghat = e.batch_norm_bwd(
             x, grad_out.contiguous(memory_format=torch.channels_last),
             y, weight, mean, invstd, ctx.relu, ctx.has_addend,
-            scale, shift)
+            scale, shift, go2)
         grad_addend = ghat if ctx.has_addend else None
         return (grad_x, grad_w, grad_b, None, None, None, None, None,
-                grad_addend, None)
+                grad_addend, None, None)
 
 
 def _bn_torch(x: torch.Tensor, bn, relu: bool,
@@ -86,9 +97,15 @@ def batch_norm(x: torch.Tensor, bn, relu: bool = False,
         xc = x.contiguous(memory_format=torch.channels_last)
         ac = None if addend is None else addend.contiguous(memory_format=torch.channels_last)
         if bn.training:
-            return _BNFunction.apply(xc, bn.weight, bn.bias, bn.running_mean,
-                                     bn.running_var, bn.momentum, bn.eps,
-                                     relu, ac, parts)
+            # block-tail BNs publish a mailbox the NEXT residual block can
+            # reroute its shortcut gradient into (ResidualGradTap)
+            cell = GradCell() if (relu and ac is not None) else None
+            out = _BNFunction.apply(xc, bn.weight, bn.bias, bn.running_mean,
+                                    bn.running_var, bn.momentum, bn.eps,
+                                    relu, ac, parts, cell)
+            if cell is not None:
+                out._amdtrain_next_cell = cell
+            return out
         return require_ext().batch_norm_fwd_eval(
             xc, bn.weight, bn.bias, bn.running_mean, bn.running_var,
             float(bn.eps), bool(relu), ac)
