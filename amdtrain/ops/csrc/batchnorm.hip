@@ -190,16 +190,22 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
 // looping 512 strided rows measured 125 us; this two-level scheme is ~2 us)
 constexpr int BN_COLLAPSE = 32;
 
+static inline int bn_collapse_slots(int nparts) {
+  // more fold-slots for large partial sets (conv-stats partials reach
+  // ~12k rows at layer1 b512; 32 slots left the collapse latency-bound)
+  return nparts >= 2048 ? 4 * BN_COLLAPSE : BN_COLLAPSE;
+}
+
 __global__ void bn_collapse_partials_kernel(const float* __restrict__ in,
                                             float* __restrict__ out,
-                                            int nparts, int twoC) {
-  // grid: BN_COLLAPSE x ceil(twoC/256); block b sums rows [b].. step 32
-  const int slot = blockIdx.x % BN_COLLAPSE;
-  const int cblk = blockIdx.x / BN_COLLAPSE;
+                                            int nparts, int twoC, int slots) {
+  // grid: slots x ceil(twoC/256); block b sums rows [b].. step slots
+  const int slot = blockIdx.x % slots;
+  const int cblk = blockIdx.x / slots;
   const int c = cblk * AMD_TPB + threadIdx.x;
   if (c >= twoC) return;
   float acc = 0.f;
-  for (int p = slot; p < nparts; p += BN_COLLAPSE)
+  for (int p = slot; p < nparts; p += slots)
     acc += in[(long)p * twoC + c];
   out[(long)slot * twoC + c] = acc;
 }
@@ -440,17 +446,18 @@ std::vector<at::Tensor> batch_norm_fwd_train(
                                         nullptr, nullptr, nullptr, nullptr,
                                         sums.data_ptr<float>(), R, (int)C);
     CHECK_CUDA_OK();
-    auto sums2 = at::empty({BN_COLLAPSE, 2 * C}, opts);
+    const int slots = bn_collapse_slots(rgrid);
+    auto sums2 = at::empty({slots, 2 * C}, opts);
     {
-      int cgrid = BN_COLLAPSE * (int)((2 * C + AMD_TPB - 1) / AMD_TPB);
+      int cgrid = slots * (int)((2 * C + AMD_TPB - 1) / AMD_TPB);
       bn_collapse_partials_kernel<<<cgrid, AMD_TPB, 0, stream>>>(
           sums.data_ptr<float>(), sums2.data_ptr<float>(), rgrid,
-          (int)(2 * C));
+          (int)(2 * C), slots);
       CHECK_CUDA_OK();
     }
     int fgrid = (int)((C + AMD_TPB - 1) / AMD_TPB);
     bn_finalize_train_kernel<<<fgrid, AMD_TPB, 0, stream>>>(
-        sums2.data_ptr<float>(), BN_COLLAPSE, weight.data_ptr<float>(),
+        sums2.data_ptr<float>(), slots, weight.data_ptr<float>(),
         bias.data_ptr<float>(), running_mean.data_ptr<float>(),
         running_var.data_ptr<float>(), mean.data_ptr<float>(),
         invstd.data_ptr<float>(), scale.data_ptr<float>(),
@@ -496,17 +503,18 @@ std::vector<at::Tensor> batch_norm_fwd_train_from_parts(
   auto y = at::empty_like(x);
   auto stream = at::cuda::getCurrentCUDAStream();
 
-  auto sums2 = at::empty({BN_COLLAPSE, 2 * C}, opts);
+  const int slots = bn_collapse_slots(nparts);
+  auto sums2 = at::empty({slots, 2 * C}, opts);
   {
-    int cgrid = BN_COLLAPSE * (int)((2 * C + AMD_TPB - 1) / AMD_TPB);
+    int cgrid = slots * (int)((2 * C + AMD_TPB - 1) / AMD_TPB);
     bn_collapse_partials_kernel<<<cgrid, AMD_TPB, 0, stream>>>(
         parts.data_ptr<float>(), sums2.data_ptr<float>(), nparts,
-        (int)(2 * C));
+        (int)(2 * C), slots);
     CHECK_CUDA_OK();
   }
   int fgrid = (int)((C + AMD_TPB - 1) / AMD_TPB);
   bn_finalize_train_kernel<<<fgrid, AMD_TPB, 0, stream>>>(
-      sums2.data_ptr<float>(), BN_COLLAPSE, weight.data_ptr<float>(),
+      sums2.data_ptr<float>(), slots, weight.data_ptr<float>(),
       bias.data_ptr<float>(), running_mean.data_ptr<float>(),
       running_var.data_ptr<float>(), mean.data_ptr<float>(),
       invstd.data_ptr<float>(), scale.data_ptr<float>(),
@@ -639,17 +647,18 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
     else REDUCE(0, false, false);
 #undef REDUCE
     CHECK_CUDA_OK();
-    auto sums2 = at::empty({BN_COLLAPSE, 2 * C}, opts);
+    const int slots = bn_collapse_slots(rgrid);
+    auto sums2 = at::empty({slots, 2 * C}, opts);
     {
-      int cgrid = BN_COLLAPSE * (int)((2 * C + AMD_TPB - 1) / AMD_TPB);
+      int cgrid = slots * (int)((2 * C + AMD_TPB - 1) / AMD_TPB);
       bn_collapse_partials_kernel<<<cgrid, AMD_TPB, 0, stream>>>(
           sums.data_ptr<float>(), sums2.data_ptr<float>(), rgrid,
-          (int)(2 * C));
+          (int)(2 * C), slots);
       CHECK_CUDA_OK();
     }
     int fgrid = (int)((C + AMD_TPB - 1) / AMD_TPB);
     bn_bwd_finalize_kernel<<<fgrid, AMD_TPB, 0, stream>>>(
-        sums2.data_ptr<float>(), BN_COLLAPSE, weight.data_ptr<float>(),
+        sums2.data_ptr<float>(), slots, weight.data_ptr<float>(),
         mean.data_ptr<float>(), invstd.data_ptr<float>(),
         gw.data_ptr<float>(), gb.data_ptr<float>(), A.data_ptr<float>(),
         Bc.data_ptr<float>(), Dc.data_ptr<float>(), R, (int)C);

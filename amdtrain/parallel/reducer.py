@@ -64,6 +64,11 @@ class BucketedReducer:
         self._grad_accum = False  # no_sync mode
         self._cb_queued = False
         self._launch_order: List[_Bucket] = []
+        # observability: how many bucket all-reduces were ENQUEUED while
+        # backward was still running (vs flushed by the final callback) —
+        # the compute/communication overlap the design promises.  Readable
+        # after each backward; asserted by tests/test_reducer.py.
+        self.last_overlap_launches = 0
         self._build_buckets(bucket_cap_mb)
         self._attach_hooks()
 
@@ -125,10 +130,13 @@ class BucketedReducer:
             torch.autograd.Variable._execution_engine.queue_callback(
                 self._final_callback)
             self._cb_queued = True
+            self.last_overlap_launches = 0
         b = self.param_to_bucket[p]
         b.ready += 1
         if b.ready == len(b.params):
             self._launch(b)
+            if b.work is not None:
+                self.last_overlap_launches += 1
 
     def _world(self) -> int:
         if not (dist.is_available() and dist.is_initialized()):

@@ -117,3 +117,41 @@ def test_reducer_four_ranks():
     for rank in range(4):
         for got, want in zip(res[rank], expected):
             assert torch.allclose(got, want, atol=1e-6)
+
+
+def test_allreduce_overlaps_backward():
+    """The reducer must ENQUEUE bucket all-reduces while backward is still
+    running (not flush everything in the final callback) — the overlap is
+    the reducer's whole point (SURVEY §5 comm plan item ii)."""
+    import torch.multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    ps = [ctx.Process(target=_overlap_worker, args=(r, 2, q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    for p in ps:
+        p.join(180)
+    results = [q.get() for _ in range(2)]
+    for nbuckets, overlapped in results:
+        assert nbuckets >= 3  # tiny cap must split the model
+        # every bucket except possibly the last (flushed by the callback)
+        # launched during backward
+        assert overlapped >= nbuckets - 1, (nbuckets, overlapped)
+
+
+def _overlap_worker(rank, world, q):
+    import os
+    import torch.distributed as dist
+    from amdtrain.parallel import NativeDDP
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29655")
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    torch.manual_seed(0)
+    m = torch.nn.Sequential(*[torch.nn.Linear(256, 256) for _ in range(8)])
+    ddp = NativeDDP(m, bucket_cap_mb=0.5)  # ~0.26 MB per layer -> ~4 buckets
+    x = torch.randn(4, 256)
+    ddp(x).sum().backward()
+    q.put((len(ddp.reducer.buckets), ddp.reducer.last_overlap_launches))
+    dist.destroy_process_group()
