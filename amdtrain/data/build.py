@@ -33,10 +33,15 @@ def build_datasets(args):
         val = SyntheticImageNet(n_val, image_size=image_size, seed=2)
         return train, val
     gpu_norm = getattr(args, "gpu_normalize", True)
+    img = getattr(args, "image_size", 224)
+    # reference crop/resize ratio (224/256, distributed.py:182-189) scales
+    # with --image-size
+    resize = max(img + 1, int(round(img * 256 / 224)))
     train = ImageFolder(os.path.join(args.data, "train"),
-                        train_transforms(gpu_normalize=gpu_norm))
+                        train_transforms(size=img, gpu_normalize=gpu_norm))
     val = ImageFolder(os.path.join(args.data, "val"),
-                      val_transforms(gpu_normalize=gpu_norm))
+                      val_transforms(size=img, resize=resize,
+                                     gpu_normalize=gpu_norm))
     return train, val
 
 
