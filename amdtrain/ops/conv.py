@@ -206,8 +206,8 @@ def conv3x3_mfma(x: torch.Tensor, weight: torch.Tensor,
 
 class _ConvGeneric(torch.autograd.Function):
     """Generic implicit-GEMM conv (element-gather im2col) — serves the 7x7
-    stem.  Input gradient is not implemented (the stem input is data); the
-    dispatcher guards on ``x.requires_grad``."""
+    stem and any other odd configuration.  Input gradient gathers dY
+    through the transposed-conv map (conv_generic_dgrad) when needed."""
 
     @staticmethod
     @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
@@ -225,7 +225,7 @@ class _ConvGeneric(torch.autograd.Function):
         if kpad != k:
             w2 = torch.nn.functional.pad(w2, (0, kpad - k))
         y2d = e.conv_generic_fwd(x2d, n, h, w, kh, kw, stride, pad, w2)
-        ctx.save_for_backward(x2d)
+        ctx.save_for_backward(x2d, weight)
         ctx.meta = (n, cin, h, w, kh, kw, stride, pad, cout, k)
         ho = (h + 2 * pad - kh) // stride + 1
         wo = (w + 2 * pad - kw) // stride + 1
@@ -235,10 +235,22 @@ class _ConvGeneric(torch.autograd.Function):
     @torch.amp.custom_bwd(device_type="cuda")
     def backward(ctx, grad_y: torch.Tensor):
         e = require_ext()
-        (x2d,) = ctx.saved_tensors
+        x2d, weight = ctx.saved_tensors
         n, cin, h, w, kh, kw, stride, pad, cout, k = ctx.meta
         gy2d = _rows(grad_y.contiguous(memory_format=torch.channels_last)) \
             .to(torch.bfloat16)
+        dx = None
+        if ctx.needs_input_grad[0]:
+            # weight permuted [Cin, KH*KW*Cout] + K2 padding
+            k2 = kh * kw * cout
+            k2p = (k2 + 31) // 32 * 32
+            w2p = weight.to(torch.bfloat16).permute(1, 2, 3, 0) \
+                .reshape(cin, k2)
+            if k2p != k2:
+                w2p = torch.nn.functional.pad(w2p, (0, k2p - k2))
+            dx2d = e.conv_generic_dgrad(gy2d, w2p.contiguous(), n, h, w,
+                                        kh, kw, stride, pad)
+            dx = dx2d.view(n, h, w, cin).permute(0, 3, 1, 2)
         if _wgrad2_enabled() and cout % 8 == 0:
             # tap-gather TN core; channels padded to a multiple of 8 so a
             # 16 B staging unit never spans taps (the stem's Cin=3 -> 8)
@@ -255,7 +267,7 @@ class _ConvGeneric(torch.autograd.Function):
                                        pad)
             dw = dw2[:, :k].view(cout, kh, kw, cin).permute(0, 3, 1, 2) \
                 .contiguous(memory_format=torch.channels_last)
-        return None, dw, None, None
+        return dx, dw, None, None
 
 
 def conv_stem_mfma(x: torch.Tensor, weight: torch.Tensor, stride: int,
@@ -299,12 +311,13 @@ class AmdConv2d(nn.Conv2d):
                     and os.environ.get("AMDTRAIN_CONV3X3", "custom")
                     == "custom"):
                 return conv3x3_mfma(x, self.weight, self.stride[0])
-            if (not x.requires_grad and self.out_channels % 16 == 0
+            if (self.out_channels % 16 == 0
                     and self.stride[0] == self.stride[1]
                     and self.padding[0] == self.padding[1]
                     and os.environ.get("AMDTRAIN_CONVSTEM", "custom")
                     == "custom"):
-                # generic element-gather path (the 7x7 stem; no input grad)
+                # generic element-gather path (stem + odd shapes; full
+                # fwd/dgrad/wgrad coverage)
                 return conv_stem_mfma(x, self.weight, self.stride[0],
                                       self.padding[0])
         return super().forward(x)

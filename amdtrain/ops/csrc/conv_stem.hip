@@ -12,7 +12,12 @@
 //
 //   fwd:   Y[m, cout] = sum_col im2col[m, col] * W2[cout, col]
 //   wgrad: dW2[cout, col] = sum_m dY[m, cout] * im2col[m, col]
-// (input dgrad is not needed: the stem input is data.)
+//   dgrad: dX[m=(n,h,w), ci] = sum_col2 im2colT[m, col2] * W2p[ci, col2]
+//          with col2 = tap*Cout + cout, im2colT gathering dY through the
+//          transposed-conv map ho = (h + pad - kh) / stride (exact
+//          divisions only), W2p = weight permuted to [Cin, KH*KW*Cout].
+//          (The stem itself never needs it — x is data — but any other
+//          odd-shaped conv does)
 #include "common.h"
 
 namespace {
@@ -38,9 +43,18 @@ struct StemCoord {
   int hb, wb;
 };
 
+template <bool DGRAD = false>
 __device__ __forceinline__ StemCoord stem_decode(long m, const StemGeom& g) {
   StemCoord u;
   long t = m;
+  if (DGRAD) {  // m ranges over INPUT pixels; gather reads dY [Hout,Wout]
+    const int w = (int)(t % g.W); t /= g.W;
+    const int h = (int)(t % g.H); t /= g.H;
+    u.n_off = t * (long)g.Hout * g.Wout;
+    u.hb = h + g.pad;
+    u.wb = w + g.pad;
+    return u;
+  }
   const int wo = (int)(t % g.Wout); t /= g.Wout;
   const int ho = (int)(t % g.Hout); t /= g.Hout;
   u.n_off = t * (long)g.H * g.W;
@@ -60,13 +74,25 @@ __device__ __forceinline__ void col_decode(int col, const StemGeom& g,
   kw = tap - kh * g.KW;
 }
 
-// element gather using hoisted row coords
+// element gather using hoisted row coords (DGRAD: g.Cin holds the conv's
+// Cout — the channel count of the gathered dY rows)
+template <bool DGRAD = false>
 __device__ __forceinline__ long stem_elem(const StemCoord& u, int col,
                                           const StemGeom& g, float inv_cin,
                                           float inv_kw) {
   if (col >= g.KH * g.KW * g.Cin) return -1;  // K padding
   int kh, kw, ci;
   col_decode(col, g, inv_cin, inv_kw, kh, kw, ci);
+  if (DGRAD) {
+    int ho2 = u.hb - kh, wo2 = u.wb - kw;
+    if (g.stride > 1) {
+      if (ho2 % g.stride || wo2 % g.stride) return -1;
+      ho2 /= g.stride;
+      wo2 /= g.stride;
+    }
+    if (ho2 < 0 || ho2 >= g.Hout || wo2 < 0 || wo2 >= g.Wout) return -1;
+    return (u.n_off + (long)ho2 * g.Wout + wo2) * g.Cin + ci;
+  }
   const int h = u.hb + kh, w = u.wb + kw;
   if (h < 0 || h >= g.H || w < 0 || w >= g.W) return -1;
   return (u.n_off + (long)h * g.W + w) * g.Cin + ci;
@@ -74,6 +100,7 @@ __device__ __forceinline__ long stem_elem(const StemCoord& u, int col,
 
 // stage a [128 m][32 col] im2col tile into LDS (per-element gather with
 // hoisted per-row decode — uc[] is precomputed once per kernel)
+template <bool DGRAD = false>
 __device__ __forceinline__ void stage_im2col(
     const bf16* __restrict__ x, const StemCoord* uc, const bool* valid,
     int c0, const StemGeom& g, float inv_cin, float inv_kw, bf16* lds) {
@@ -85,8 +112,9 @@ __device__ __forceinline__ void stage_im2col(
     bf16 v[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      long idx = valid[rnd] ? stem_elem(uc[rnd], cc0 + j, g, inv_cin, inv_kw)
-                            : -1;
+      long idx = valid[rnd]
+                     ? stem_elem<DGRAD>(uc[rnd], cc0 + j, g, inv_cin, inv_kw)
+                     : -1;
       v[j] = idx < 0 ? bf16(0.f) : x[idx];
     }
     *(Pack<bf16, 8>*)(lds + unit * 8) = *(Pack<bf16, 8>*)v;
@@ -110,6 +138,9 @@ __device__ __forceinline__ void stage_rows(
   }
 }
 
+// DGRAD instantiation computes dX[m, cin] from gathered dY rows and the
+// permuted weight; operand roles mirror fwd exactly.
+template <bool DGRAD = false>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 conv_generic_fwd_kernel(const bf16* __restrict__ x,
                         const bf16* __restrict__ W2, bf16* __restrict__ Y,
@@ -137,12 +168,12 @@ conv_generic_fwd_kernel(const bf16* __restrict__ x,
   for (int rnd = 0; rnd < 2; ++rnd) {
     long m = m0 + (((rnd * GEMM_TPB) + t) >> 2);
     valid[rnd] = m < M;
-    uc[rnd] = stem_decode(valid[rnd] ? m : M - 1, g);
+    uc[rnd] = stem_decode<DGRAD>(valid[rnd] ? m : M - 1, g);
   }
 
   for (int ks = 0; ks < g.Kpad / BK; ++ks) {
     __syncthreads();
-    stage_im2col(x, uc, valid, ks * BK, g, inv_cin, inv_kw, As);
+    stage_im2col<DGRAD>(x, uc, valid, ks * BK, g, inv_cin, inv_kw, As);
     stage_rows(W2, g.Kpad, n0, Cout, ks * BK, Bs);
     __syncthreads();
     bf16x8 a[4], b[4];
@@ -313,11 +344,33 @@ at::Tensor conv_generic_fwd(at::Tensor x2d, long Nn, long H, long W,
   auto y = at::empty({M, Cout}, x2d.options());
   int nbm = (int)((M + 127) / 128), nbn = (int)((Cout + 127) / 128);
   auto stream = at::cuda::getCurrentCUDAStream();
-  conv_generic_fwd_kernel<<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+  conv_generic_fwd_kernel<false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
       (const bf16*)x2d.const_data_ptr(), (const bf16*)w2.const_data_ptr(),
       (bf16*)y.data_ptr(), M, (int)Cout, g, nbm, nbn);
   CHECK_CUDA_OK();
   return y;
+}
+
+// dy2d: [N*Hout*Wout, Cout]; w2p: [Cin, pad32(KH*KW*Cout)] (weight permuted
+// to cin-major, host-padded).  Returns dx2d [N*H*W, Cin].
+at::Tensor conv_generic_dgrad(at::Tensor dy2d, at::Tensor w2p, long Nn,
+                              long H, long W, long KH, long KW, long stride,
+                              long pad) {
+  TORCH_CHECK(dy2d.is_cuda() && dy2d.scalar_type() == at::kBFloat16);
+  long Cout = dy2d.size(1), Cin = w2p.size(0);
+  // geometry carries the FORWARD conv dims; g.Cin = Cout (gathered rows)
+  auto g = make_geom(H, W, KH, KW, Cout, stride, pad);
+  TORCH_CHECK(w2p.size(1) == g.Kpad, "w2p must be K-padded to ", g.Kpad);
+  TORCH_CHECK(dy2d.size(0) == Nn * g.Hout * g.Wout);
+  long M = Nn * H * W;
+  auto dx = at::empty({M, Cin}, dy2d.options());
+  int nbm = (int)((M + 127) / 128), nbn = (int)((Cin + 127) / 128);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  conv_generic_fwd_kernel<true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+      (const bf16*)dy2d.const_data_ptr(), (const bf16*)w2p.const_data_ptr(),
+      (bf16*)dx.data_ptr(), M, (int)Cin, g, nbm, nbn);
+  CHECK_CUDA_OK();
+  return dx;
 }
 
 at::Tensor conv_generic_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn,

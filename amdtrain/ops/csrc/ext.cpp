@@ -89,6 +89,9 @@ at::Tensor conv_generic_fwd(at::Tensor x2d, long Nn, long H, long W, long KH,
 at::Tensor conv_generic_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn,
                               long H, long W, long KH, long KW, long stride,
                               long pad);
+at::Tensor conv_generic_dgrad(at::Tensor dy2d, at::Tensor w2p, long Nn,
+                              long H, long W, long KH, long KW, long stride,
+                              long pad);
 
 // pool.hip
 std::vector<at::Tensor> max_pool_3x3_s2_fwd(at::Tensor x);
@@ -148,6 +151,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tr16_probe", &tr16_probe);
   m.def("conv_generic_fwd", &conv_generic_fwd);
   m.def("conv_generic_wgrad", &conv_generic_wgrad);
+  m.def("conv_generic_dgrad", &conv_generic_dgrad);
   m.def("max_pool_3x3_s2_fwd", &max_pool_3x3_s2_fwd);
   m.def("max_pool_3x3_s2_bwd", &max_pool_3x3_s2_bwd);
   m.def("global_avg_pool_fwd", &global_avg_pool_fwd);

@@ -380,3 +380,31 @@ def test_deterministic_wgrad(monkeypatch):
     a = _C.conv3x3_wgrad(gy2d, x2d, 2, 14, 14, 1)
     b = _C.conv3x3_wgrad(gy2d, x2d, 2, 14, 14, 1)
     assert torch.equal(a, b)
+
+
+@pytest.mark.parametrize("cfg", [
+    (3, 16, 7, 2, 3, 22),    # stem-like (Cin=3, 7x7 s2 p3)
+    (32, 48, 5, 1, 2, 17),   # odd 5x5
+    (16, 32, 7, 3, 2, 21),   # stride 3
+])
+def test_conv_generic_dgrad(cfg):
+    """Generic-conv input gradient vs fp32 torch."""
+    cin, cout, k, s, p, hw = cfg
+    from amdtrain.ops.conv import conv_stem_mfma
+    torch.manual_seed(0)
+    x = torch.randn(2, cin, hw, hw, device=DEV) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    w = torch.randn(cout, cin, k, k, device=DEV) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        y = conv_stem_mfma(x, w, s, p)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    xr = x.detach().clone().float().requires_grad_(True)
+    wr = w.detach().clone().float().requires_grad_(True)
+    yr = F.conv2d(xr, wr, stride=s, padding=p)
+    yr.backward(gy.float())
+    assert torch.allclose(x.grad.float(), xr.grad, atol=0.5, rtol=0.05), \
+        (x.grad.float() - xr.grad).abs().max().item()
+    assert torch.allclose(w.grad.float(), wr.grad, atol=1.0, rtol=0.05), \
+        (w.grad.float() - wr.grad).abs().max().item()
