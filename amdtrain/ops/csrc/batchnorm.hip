@@ -68,32 +68,18 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
         }
       }
     }
-    // 2-way row unroll: two independent row streams in flight per thread
-    // (the bwd reduce is 3-5 HBM streams and was latency-bound at 1)
-    for (long r = r0 + phase; r < r1; r += 2 * rstep) {
-      const bool has2 = r + rstep < r1;
-      const long rB = has2 ? r + rstep : r;  // clamp: loads stay in bounds
+    for (long r = r0 + phase; r < r1; r += rstep) {
       const long base = r * C + c0;
-      const long base2 = rB * C + c0;
       Pack<T, VEC> xv = *(const Pack<T, VEC>*)(x + base);
-      Pack<T, VEC> xw = *(const Pack<T, VEC>*)(x + base2);
-      Pack<T, VEC> gv, gv2, yv, gh, gw_, gw2, yw, gh2;
+      Pack<T, VEC> gv, gv2, yv, gh;
       if (BWD) {
         gv = *(const Pack<T, VEC>*)(go + base);
-        gw_ = *(const Pack<T, VEC>*)(go + base2);
-        if (GO2) {
-          gv2 = *(const Pack<T, VEC>*)(goB + base);
-          gw2 = *(const Pack<T, VEC>*)(goB + base2);
-        }
-        if (MASK == 1) {
-          yv = *(const Pack<T, VEC>*)(y + base);
-          yw = *(const Pack<T, VEC>*)(y + base2);
-        }
+        if (GO2) gv2 = *(const Pack<T, VEC>*)(goB + base);
+        if (MASK == 1) yv = *(const Pack<T, VEC>*)(y + base);
       }
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
         float xe = to_f32(xv.v[k]);
-        float xe2 = to_f32(xw.v[k]);
         if (BWD) {
           float ge = to_f32(gv.v[k]);
           if (GO2) ge += to_f32(gv2.v[k]);
@@ -102,28 +88,12 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
           sa[k] += ge;
           sb[k] += ge * (xe - mk[k]) * ik[k];
           if (WG) gh.v[k] = from_f32<T>(ge);
-          float ge2 = to_f32(gw_.v[k]);
-          if (GO2) ge2 += to_f32(gw2.v[k]);
-          if (MASK == 1 && to_f32(yw.v[k]) <= 0.f) ge2 = 0.f;
-          if (MASK == 2 && sck[k] * xe2 + shk[k] <= 0.f) ge2 = 0.f;
-          if (has2) {
-            sa[k] += ge2;
-            sb[k] += ge2 * (xe2 - mk[k]) * ik[k];
-          }
-          if (WG) gh2.v[k] = from_f32<T>(ge2);
         } else {
           sa[k] += xe;
           sb[k] += xe * xe;
-          if (has2) {
-            sa[k] += xe2;
-            sb[k] += xe2 * xe2;
-          }
         }
       }
-      if (BWD && WG) {
-        *(Pack<T, VEC>*)(ghat_out + base) = gh;
-        if (has2) *(Pack<T, VEC>*)(ghat_out + base2) = gh2;
-      }
+      if (BWD && WG) *(Pack<T, VEC>*)(ghat_out + base) = gh;
     }
     // LDS reduce across phases, then one global atomicAdd per channel
     __shared__ float lds[AMD_TPB * VEC];
