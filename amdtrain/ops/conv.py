@@ -218,15 +218,24 @@ class _ConvGeneric(torch.autograd.Function):
         cout, _, kh, kw = weight.shape
         xc = x.contiguous(memory_format=torch.channels_last)
         x2d = _rows(xc)
-        k = kh * kw * cin
+        w4 = weight.contiguous(memory_format=torch.channels_last) \
+            .permute(0, 2, 3, 1)                       # [Cout, KH, KW, Cin]
+        if cin < 8:
+            # stem fast path: pad channels to 8 so every 16 B staging unit
+            # is one tap — the kernel's vectorized tap-gather (C8) route
+            x2d = torch.nn.functional.pad(x2d, (0, 8 - cin))
+            w4 = torch.nn.functional.pad(w4, (0, 8 - cin))
+            cin_k = 8
+        else:
+            cin_k = cin
+        k = kh * kw * cin_k
         kpad = (k + 31) // 32 * 32
-        w2 = weight.contiguous(memory_format=torch.channels_last) \
-            .permute(0, 2, 3, 1).reshape(cout, k)
+        w2 = w4.reshape(cout, k)
         if kpad != k:
             w2 = torch.nn.functional.pad(w2, (0, kpad - k))
         y2d = e.conv_generic_fwd(x2d, n, h, w, kh, kw, stride, pad, w2)
         ctx.save_for_backward(x2d, weight)
-        ctx.meta = (n, cin, h, w, kh, kw, stride, pad, cout, k)
+        ctx.meta = (n, cin, h, w, kh, kw, stride, pad, cout, k, cin_k)
         ho = (h + 2 * pad - kh) // stride + 1
         wo = (w + 2 * pad - kw) // stride + 1
         return y2d.view(n, ho, wo, cout).permute(0, 3, 1, 2)
@@ -236,7 +245,7 @@ class _ConvGeneric(torch.autograd.Function):
     def backward(ctx, grad_y: torch.Tensor):
         e = require_ext()
         x2d, weight = ctx.saved_tensors
-        n, cin, h, w, kh, kw, stride, pad, cout, k = ctx.meta
+        n, cin, h, w, kh, kw, stride, pad, cout, k, cin_k = ctx.meta
         gy2d = _rows(grad_y.contiguous(memory_format=torch.channels_last)) \
             .to(torch.bfloat16)
         dx = None
@@ -252,20 +261,17 @@ class _ConvGeneric(torch.autograd.Function):
                                         kh, kw, stride, pad)
             dx = dx2d.view(n, h, w, cin).permute(0, 3, 1, 2)
         if _wgrad2_enabled() and cout % 8 == 0:
-            # tap-gather TN core; channels padded to a multiple of 8 so a
-            # 16 B staging unit never spans taps (the stem's Cin=3 -> 8)
-            cinp = (cin + 7) // 8 * 8
-            xp = x2d if cinp == cin else \
-                torch.nn.functional.pad(x2d, (0, cinp - cin))
-            dwp = e.tn2_wgrad(gy2d, xp, kh * kw, n, h, w, stride, 2,
-                              kh, kw, pad)                 # [Cout, taps*cinp]
-            dw = dwp.view(cout, kh * kw, cinp)[:, :, :cin] \
+            # tap-gather TN core; x2d is already channel-padded (fwd)
+            dwp = e.tn2_wgrad(gy2d, x2d, kh * kw, n, h, w, stride, 2,
+                              kh, kw, pad)               # [Cout, taps*cin_k]
+            dw = dwp.view(cout, kh * kw, cin_k)[:, :, :cin] \
                 .view(cout, kh, kw, cin).permute(0, 3, 1, 2) \
                 .contiguous(memory_format=torch.channels_last)
         else:
             dw2 = e.conv_generic_wgrad(gy2d, x2d, n, h, w, kh, kw, stride,
                                        pad)
-            dw = dw2[:, :k].view(cout, kh, kw, cin).permute(0, 3, 1, 2) \
+            dw = dw2[:, :k].view(cout, kh * kw, cin_k)[:, :, :cin] \
+                .reshape(cout, kh, kw, cin).permute(0, 3, 1, 2) \
                 .contiguous(memory_format=torch.channels_last)
         return dx, dw, None, None
 

@@ -138,13 +138,42 @@ __device__ __forceinline__ void stage_rows(
   }
 }
 
+
+// Cin==8 fast path (the channel-padded stem): one 16 B unit == one tap's
+// 8 channels, so the A tile stages via global_load_lds with a per-unit
+// tap-gathered source row — same structure as conv3x3.hip's stage, no
+// per-element scalar gather (the r1 elem-gather fwd measured 1.2 ms/step
+// at b512, VALU-bound).
+__device__ __forceinline__ void stage_im2col8(
+    const bf16* __restrict__ x, const StemCoord* uc, const bool* valid,
+    int c0, const StemGeom& g, const bf16* __restrict__ zp, bf16* lds) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int rnd = 0; rnd < 2; ++rnd) {
+    const int unit = rnd * GEMM_TPB + t;  // 4 units (taps) per m-row
+    const int tap = (c0 >> 3) + (unit & 3);
+    const bf16* src = zp;
+    if (valid[rnd] && tap < g.KH * g.KW) {
+      const int kh = tap / g.KW, kw = tap - (tap / g.KW) * g.KW;
+      const int h = uc[rnd].hb + kh, w = uc[rnd].wb + kw;
+      if (h >= 0 && h < g.H && w >= 0 && w < g.W)
+        src = x + (uc[rnd].n_off + (long)h * g.W + w) * 8;
+    }
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)(lds + unit * 8), 16,
+        0, 0);
+  }
+}
+
 // DGRAD instantiation computes dX[m, cin] from gathered dY rows and the
 // permuted weight; operand roles mirror fwd exactly.
-template <bool DGRAD = false>
+template <bool DGRAD = false, bool C8 = false>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 conv_generic_fwd_kernel(const bf16* __restrict__ x,
                         const bf16* __restrict__ W2, bf16* __restrict__ Y,
-                        long M, int Cout, StemGeom g, int nbm, int nbn) {
+                        long M, int Cout, StemGeom g, int nbm, int nbn,
+                        const bf16* __restrict__ zp) {
   __shared__ bf16 As[128 * BK];
   __shared__ bf16 Bs[128 * BK];
   const int bid = blockIdx.x;
@@ -173,7 +202,10 @@ conv_generic_fwd_kernel(const bf16* __restrict__ x,
 
   for (int ks = 0; ks < g.Kpad / BK; ++ks) {
     __syncthreads();
-    stage_im2col<DGRAD>(x, uc, valid, ks * BK, g, inv_cin, inv_kw, As);
+    if (C8)
+      stage_im2col8(x, uc, valid, ks * BK, g, zp, As);
+    else
+      stage_im2col<DGRAD>(x, uc, valid, ks * BK, g, inv_cin, inv_kw, As);
     stage_rows(W2, g.Kpad, n0, Cout, ks * BK, Bs);
     __syncthreads();
     bf16x8 a[4], b[4];
@@ -330,9 +362,17 @@ static StemGeom make_geom(long H, long W, long KH, long KW, long Cin,
   return g;
 }
 
+static at::Tensor stem_zero_page(const at::Tensor& like) {
+  static thread_local at::Tensor zp;
+  if (!zp.defined() || zp.device() != like.device())
+    zp = at::zeros({16}, like.options().dtype(at::kBFloat16));
+  return zp;
+}
+
 }  // namespace
 
-// x2d: [N*H*W, Cin] bf16 NHWC rows; w2: [Cout, Kpad] (host-padded)
+// x2d: [N*H*W, Cin] bf16 NHWC rows; w2: [Cout, Kpad] (host-padded).
+// Cin == 8 (channel-padded stem) takes the vectorized tap-gather staging.
 at::Tensor conv_generic_fwd(at::Tensor x2d, long Nn, long H, long W,
                             long KH, long KW, long stride, long pad,
                             at::Tensor w2) {
@@ -344,9 +384,17 @@ at::Tensor conv_generic_fwd(at::Tensor x2d, long Nn, long H, long W,
   auto y = at::empty({M, Cout}, x2d.options());
   int nbm = (int)((M + 127) / 128), nbn = (int)((Cout + 127) / 128);
   auto stream = at::cuda::getCurrentCUDAStream();
-  conv_generic_fwd_kernel<false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
-      (const bf16*)x2d.const_data_ptr(), (const bf16*)w2.const_data_ptr(),
-      (bf16*)y.data_ptr(), M, (int)Cout, g, nbm, nbn);
+  auto zp = stem_zero_page(x2d);
+  if (Cin == 8 && g.Kpad == (KH * KW * 8 + 31) / 32 * 32)
+    conv_generic_fwd_kernel<false, true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+        (const bf16*)x2d.const_data_ptr(), (const bf16*)w2.const_data_ptr(),
+        (bf16*)y.data_ptr(), M, (int)Cout, g, nbm, nbn,
+        (const bf16*)zp.const_data_ptr());
+  else
+    conv_generic_fwd_kernel<false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+        (const bf16*)x2d.const_data_ptr(), (const bf16*)w2.const_data_ptr(),
+        (bf16*)y.data_ptr(), M, (int)Cout, g, nbm, nbn,
+        (const bf16*)zp.const_data_ptr());
   CHECK_CUDA_OK();
   return y;
 }
@@ -366,9 +414,11 @@ at::Tensor conv_generic_dgrad(at::Tensor dy2d, at::Tensor w2p, long Nn,
   auto dx = at::empty({M, Cin}, dy2d.options());
   int nbm = (int)((M + 127) / 128), nbn = (int)((Cin + 127) / 128);
   auto stream = at::cuda::getCurrentCUDAStream();
+  auto zp = stem_zero_page(dy2d);
   conv_generic_fwd_kernel<true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
       (const bf16*)dy2d.const_data_ptr(), (const bf16*)w2p.const_data_ptr(),
-      (bf16*)dx.data_ptr(), M, (int)Cin, g, nbm, nbn);
+      (bf16*)dx.data_ptr(), M, (int)Cin, g, nbm, nbn,
+      (const bf16*)zp.const_data_ptr());
   CHECK_CUDA_OK();
   return dx;
 }
