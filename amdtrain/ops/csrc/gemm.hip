@@ -242,6 +242,96 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                       fr, fq, nbm, As);
 }
 
+// ---- dual-N-tile BT GEMM (A-panel reuse) ---------------------------------
+// The wide-N, small-K conv shapes (N>=256, K<=256: the bottleneck expand
+// convs) measured AT the re-read bandwidth floor: each 128-col block of
+// tiles re-fetches the full A panel (tools/bench_conv1x1.py).  This variant
+// computes TWO n-tiles per block from ONE staged A tile, halving A traffic.
+// acc grows to 4x8 fragments (128 VGPR) — still spill-free (checked in the
+// build log).
+__global__ void __launch_bounds__(GEMM_TPB, 2)
+gemm_bt_n2_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
+                  bf16* __restrict__ C, long M, long N, long K, int nbm,
+                  int nbn2, float* __restrict__ stats, int nbm_stats) {
+  __shared__ bf16 As[BM * BK];
+  __shared__ bf16 Bs[256 * BK];
+
+  const int bid = xcd_swizzle(blockIdx.x, nbm * nbn2);
+  const int bm = bid / nbn2, bn = bid % nbn2;
+  const long m0 = (long)bm * BM, n0 = (long)bn * 256;
+
+  const int t = threadIdx.x;
+  const int wave = t / AMD_WAVE;
+  const int lane = t % AMD_WAVE;
+  const int wm = (wave >> 1) * 64;
+  const int wn = (wave & 1) * 64;
+  const int fr = lane & 15;
+  const int fq = lane >> 4;
+
+  f32x4 acc[4][8];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const long ksteps = K / BK;
+  for (long kt = 0; kt < ksteps; ++kt) {
+    __syncthreads();
+    stage_tile_128x32(A, K, m0, M, kt * BK, As);
+    stage_tile_128x32(B, K, n0, N, kt * BK, Bs);
+    stage_tile_128x32(B, K, n0 + 128, N, kt * BK, Bs + 128 * BK);
+    __syncthreads();
+
+    bf16x8 a[4], b[8];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      a[i] = *(const bf16x8*)&As[(wm + i * 16 + fr) * BK + fq * 8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int jn = (j < 4) ? wn + j * 16 : 128 + wn + (j - 4) * 16;
+      b[j] = *(const bf16x8*)&Bs[jn * BK + fq * 8];
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a[i], b[j], acc[i][j], 0, 0, 0);
+  }
+
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = m0 + wm + i * 16 + fq * 4 + r;
+        long col = n0 + ((j < 4) ? wn + j * 16 : 128 + wn + (j - 4) * 16)
+                   + fr;
+        if (row < M && col < N)
+          C[row * N + col] = __float2bfloat16(acc[i][j][r]);
+      }
+  if (stats != nullptr) {
+    // per-block column (sum, sumsq) partials for the fused BN pipeline —
+    // two 128-col halves through the shared epilogue helper
+    float accl[4][4][4], acch[4][4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          accl[i][j][r] = acc[i][j][r];
+          acch[i][j][r] = acc[i][j + 4][r];
+        }
+    epilogue_stats<4>(stats, (const float*)accl, m0, M, n0, N, bm, wm, wn,
+                      fr, fq, nbm_stats, As);
+    __syncthreads();
+    epilogue_stats<4>(stats, (const float*)acch, m0, M, n0 + 128, N, bm,
+                      wm, wn, fr, fq, nbm_stats, As);
+  }
+}
+
 // ---- split-K BT GEMM for tiny-M shapes (the FC classifier) ---------------
 // At [B,2048]x[2048,1000] the plain kernel has only ceil(B/128)*8 blocks —
 // a fraction of the 256-CU chip — while K=2048 runs 64 serial K-steps.
@@ -525,6 +615,17 @@ at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out,
   }
   const long tiles = (long)nbm * nbn;
   const long ksteps = K / BK;
+  if (!f32_out && !addend && !bias && N % 256 == 0 && K <= 256 &&
+      (long)nbm * (nbn / 2) >= 256) {
+    // wide-N small-K (bottleneck expand convs): dual-n-tile variant halves
+    // the A-panel re-reads (measured at the re-read BW floor otherwise)
+    int nbn2 = nbn / 2;
+    gemm_bt_n2_kernel<<<nbm * nbn2, GEMM_TPB, 0, stream>>>(
+        (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
+        (bf16*)C.data_ptr(), M, N, K, nbm, nbn2, nullptr, nbm);
+    CHECK_CUDA_OK();
+    return C;
+  }
   if (!f32_out && tiles <= 128 && ksteps >= 2 && 512 / tiles >= 2) {
     // tiny-M path (FC classifier fwd/dgrad): split K to fill the chip
     const int ksplit = (int)std::min<long>(ksteps, 512 / tiles);
@@ -569,9 +670,17 @@ std::vector<at::Tensor> gemm_bt_stats(at::Tensor A, at::Tensor B) {
   auto stats = at::empty({nbm, 2 * N}, Ac.options().dtype(at::kFloat));
   StrideMap sm{0, 0, 0, 0, 1};
   auto stream = at::cuda::getCurrentCUDAStream();
-  gemm_bt_kernel<false, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
-      (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-      C.data_ptr(), M, N, K, nbm, nbn, sm, stats.data_ptr<float>());
+  if (N % 256 == 0 && K <= 256 && (long)nbm * (nbn / 2) >= 256) {
+    int nbn2 = nbn / 2;
+    gemm_bt_n2_kernel<<<nbm * nbn2, GEMM_TPB, 0, stream>>>(
+        (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
+        (bf16*)C.data_ptr(), M, N, K, nbm, nbn2, stats.data_ptr<float>(),
+        nbm);
+  } else {
+    gemm_bt_kernel<false, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+        (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
+        C.data_ptr(), M, N, K, nbm, nbn, sm, stats.data_ptr<float>());
+  }
   CHECK_CUDA_OK();
   return {C, stats};
 }

@@ -183,3 +183,24 @@ def test_amd_linear_parity():
     assert torch.allclose(xc.grad.float(), xr.grad, atol=2.0, rtol=0.05)
     assert torch.allclose(lin.weight.grad, wr.grad, atol=2.0, rtol=0.05)
     assert torch.allclose(lin.bias.grad, br.grad, atol=1.0, rtol=0.05)
+
+
+def test_gemm_bt_dual_n_tile():
+    """Wide-N small-K dual-n-tile variant (A-panel reuse) vs matmul, plus
+    its fused BN-stats epilogue vs direct column sums."""
+    from amdtrain import _C
+    torch.manual_seed(0)
+    M, N, K = 40960 + 96, 256, 64  # ragged M; grid large enough to engage
+    A = torch.randn(M, K, device="cuda").bfloat16()
+    B = torch.randn(N, K, device="cuda").bfloat16()
+    y = _C.gemm_bt(A, B, False)
+    ref = (A.float() @ B.float().t())
+    assert torch.allclose(y.float(), ref, atol=1.0, rtol=0.02), \
+        (y.float() - ref).abs().max().item()
+    y2, stats = _C.gemm_bt_stats(A, B)
+    assert torch.equal(y2, y)
+    s1 = stats[:, :N].sum(0)
+    s2 = stats[:, N:].sum(0)
+    yf = y.float()
+    assert torch.allclose(s1, yf.sum(0), rtol=2e-2, atol=2.0)
+    assert torch.allclose(s2, (yf * yf).sum(0), rtol=2e-2, atol=20.0)
