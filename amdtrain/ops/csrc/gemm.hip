@@ -289,7 +289,7 @@ gemm_bt_n2_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const int jn = (j < 4) ? wn + j * 16 : 128 + wn + (j - 4) * 16;
-      b[j] = *(const bf16x8*)&Bs[jn * BK + fq * 8];
+      b[j] = *(const bf16x8*)&Bs[(jn + fr) * BK + fq * 8];
     }
 #pragma unroll
     for (int i = 0; i < 4; ++i)
