@@ -244,11 +244,12 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
 
 // ---- dual-N-tile BT GEMM (A-panel reuse) ---------------------------------
 // The wide-N, small-K conv shapes (N>=256, K<=256: the bottleneck expand
-// convs) measured AT the re-read bandwidth floor: each 128-col block of
-// tiles re-fetches the full A panel (tools/bench_conv1x1.py).  This variant
-// computes TWO n-tiles per block from ONE staged A tile, halving A traffic.
-// acc grows to 4x8 fragments (128 VGPR) — still spill-free (checked in the
-// build log).
+// convs) sit at ~4.2 TB/s of min-traffic (tools/bench_conv1x1.py).  This
+// variant computes TWO n-tiles per block from ONE staged A tile, halving
+// nominal A traffic.  MEASURED NEUTRAL on those shapes (the ceiling is a
+// uniform streaming-efficiency wall, not the A re-reads — the rr-floor
+// coincidence in the first analysis was misleading); kept because it also
+// halves launch width and is covered by tests.
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 gemm_bt_n2_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                   bf16* __restrict__ C, long M, long N, long K, int nbm,
@@ -617,8 +618,8 @@ at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out,
   const long ksteps = K / BK;
   if (!f32_out && !addend && !bias && N % 256 == 0 && K <= 256 &&
       (long)nbm * (nbn / 2) >= 256) {
-    // wide-N small-K (bottleneck expand convs): dual-n-tile variant halves
-    // the A-panel re-reads (measured at the re-read BW floor otherwise)
+    // wide-N small-K: dual-n-tile variant (see kernel comment; measured
+    // perf-neutral, halves nominal A traffic and launch width)
     int nbn2 = nbn / 2;
     gemm_bt_n2_kernel<<<nbm * nbn2, GEMM_TPB, 0, stream>>>(
         (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),

@@ -202,5 +202,8 @@ def test_gemm_bt_dual_n_tile():
     s1 = stats[:, :N].sum(0)
     s2 = stats[:, N:].sum(0)
     yf = y.float()
-    assert torch.allclose(s1, yf.sum(0), rtol=2e-2, atol=2.0)
-    assert torch.allclose(s2, (yf * yf).sum(0), rtol=2e-2, atol=20.0)
+    # stats accumulate the fp32 accumulators; yf is the bf16-ROUNDED output,
+    # so the column sums random-walk apart by ~0.02*sqrt(M) (~4 at M=41k)
+    atol1 = 0.05 * (M ** 0.5)
+    assert torch.allclose(s1, yf.sum(0), rtol=2e-2, atol=atol1)
+    assert torch.allclose(s2, (yf * yf).sum(0), rtol=2e-2, atol=40 * atol1)
