@@ -31,7 +31,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--batch-per-gpu", type=int, default=2048)
+    p.add_argument("--batch-per-gpu", type=int, default=3072)  # swept: +2% over 2048, 136 GB/GPU
     p.add_argument("--arch", type=str, default="resnet50")
     p.add_argument("--dtype", type=str, default="bf16",
                    choices=["bf16", "fp32"])
