@@ -85,16 +85,17 @@ class BasicBlock(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x
         cell = None
-        if (self.downsample is None and self.training and x.is_cuda
-                and torch.is_grad_enabled() and _residual_fuse_enabled()):
+        if (self.training and x.is_cuda and torch.is_grad_enabled()
+                and _residual_fuse_enabled()):
             cell = getattr(x, "_amdtrain_next_cell", None)
         out = self.bn1.forward_relu(self.conv1(x))
         out = self.conv2(out)
-        if self.downsample is not None:
-            identity = self.downsample(x)
-        elif cell is not None:
+        if cell is not None:
             cell.armed = True
-            identity = ResidualGradTap.apply(x, cell)
+            tapped = ResidualGradTap.apply(x, cell)
+            identity = self.downsample(tapped)                 if self.downsample is not None else tapped
+        elif self.downsample is not None:
+            identity = self.downsample(x)
         return OF.bn_add_relu(out, self.bn2, identity)
 
 
@@ -121,20 +122,24 @@ class Bottleneck(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x
         cell = None
-        if (self.downsample is None and self.training and x.is_cuda
-                and torch.is_grad_enabled() and _residual_fuse_enabled()):
-            # identity-shortcut block: reroute the shortcut gradient into
-            # the PRODUCING block-tail BN's backward (mailbox attached to
-            # x by that BN) — kills the eager add at x's AccumulateGrad
+        if (self.training and x.is_cuda and torch.is_grad_enabled()
+                and _residual_fuse_enabled()):
+            # reroute the shortcut-branch gradient into the PRODUCING
+            # block-tail BN's backward (mailbox attached to x by that BN) —
+            # kills the eager add at x's AccumulateGrad.  Works for both
+            # identity shortcuts (tap feeds bn3's addend directly) and
+            # downsample shortcuts (tap feeds the downsample conv, whose
+            # backward runs before the producing BN's).
             cell = getattr(x, "_amdtrain_next_cell", None)
         out = self.bn1.forward_relu(self.conv1(x))
         out = self.bn2.forward_relu(self.conv2(out))
         out = self.conv3(out)
-        if self.downsample is not None:
-            identity = self.downsample(x)
-        elif cell is not None:
+        if cell is not None:
             cell.armed = True
-            identity = ResidualGradTap.apply(x, cell)
+            tapped = ResidualGradTap.apply(x, cell)
+            identity = self.downsample(tapped)                 if self.downsample is not None else tapped
+        elif self.downsample is not None:
+            identity = self.downsample(x)
         return OF.bn_add_relu(out, self.bn3, identity)
 
 
