@@ -25,6 +25,9 @@ namespace {
 //   MASK 0: ghat = go (no relu / pre-masked input)
 //   MASK 1: ghat = go * (y > 0)                  (block-tail BN: y has addend)
 //   MASK 2: ghat = go * (scale*x + shift > 0)    (inner BN: skip the y read)
+//   MASK 3: ghat = go * fwd-bitmask              (block-tail: 1/16 the bytes
+//                                                 of the y read; y = the
+//                                                 mask byte stream here)
 // WG: additionally materialize ghat (consumed by the mask-free bwd apply and
 // as the residual addend gradient) — turns reduce into 3r+1w so apply drops
 // to 2r+1w (guide: every pass here is an HBM-bound R x C stream).
@@ -72,10 +75,13 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
       const long base = r * C + c0;
       Pack<T, VEC> xv = *(const Pack<T, VEC>*)(x + base);
       Pack<T, VEC> gv, gv2, yv, gh;
+      unsigned int mbits = 0;
       if (BWD) {
         gv = *(const Pack<T, VEC>*)(go + base);
         if (GO2) gv2 = *(const Pack<T, VEC>*)(goB + base);
         if (MASK == 1) yv = *(const Pack<T, VEC>*)(y + base);
+        if (MASK == 3)
+          mbits = ((const unsigned char*)y)[base / VEC];
       }
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
@@ -85,6 +91,7 @@ bn_reduce_kernel(const T* __restrict__ x, const T* __restrict__ go,
           if (GO2) ge += to_f32(gv2.v[k]);
           if (MASK == 1 && to_f32(yv.v[k]) <= 0.f) ge = 0.f;
           if (MASK == 2 && sck[k] * xe + shk[k] <= 0.f) ge = 0.f;
+          if (MASK == 3 && !(mbits & (1u << k))) ge = 0.f;
           sa[k] += ge;
           sb[k] += ge * (xe - mk[k]) * ik[k];
           if (WG) gh.v[k] = from_f32<T>(ge);
@@ -258,7 +265,8 @@ __global__ void bn_finalize_eval_kernel(
 template <typename T, int VEC, bool RELU, bool HAS_ADD, bool POW2>
 __global__ void __launch_bounds__(AMD_TPB)
 bn_apply_kernel(const T* __restrict__ x, const T* __restrict__ z,
-                T* __restrict__ y, const float* __restrict__ scale,
+                T* __restrict__ y, unsigned char* __restrict__ mask,
+                const float* __restrict__ scale,
                 const float* __restrict__ shift, long total_vec, int C) {
   extern __shared__ float sm[];  // [2][C]
   const int gpr = C / VEC;
@@ -282,19 +290,30 @@ bn_apply_kernel(const T* __restrict__ x, const T* __restrict__ z,
     const int c0 = (int)(POW2 ? (i & (gpr - 1)) : (i % gpr)) * VEC;
     const int c02 = (int)(POW2 ? (i2 & (gpr - 1)) : (i2 % gpr)) * VEC;
     Pack<T, VEC> yv, yv2;
+    unsigned int mb = 0, mb2 = 0;
 #pragma unroll
     for (int k = 0; k < VEC; ++k) {
       float v = sm[c0 + k] * to_f32(xv.v[k]) + sm[C + c0 + k];
       if (HAS_ADD) v += to_f32(zv.v[k]);
-      if (RELU) v = fmaxf(v, 0.f);
+      if (RELU) {
+        if (mask != nullptr && v > 0.f) mb |= 1u << k;
+        v = fmaxf(v, 0.f);
+      }
       yv.v[k] = from_f32<T>(v);
       float v2 = sm[c02 + k] * to_f32(xv2.v[k]) + sm[C + c02 + k];
       if (HAS_ADD) v2 += to_f32(zv2.v[k]);
-      if (RELU) v2 = fmaxf(v2, 0.f);
+      if (RELU) {
+        if (mask != nullptr && v2 > 0.f) mb2 |= 1u << k;
+        v2 = fmaxf(v2, 0.f);
+      }
       yv2.v[k] = from_f32<T>(v2);
     }
     *(Pack<T, VEC>*)(y + i * VEC) = yv;
     if (has2) *(Pack<T, VEC>*)(y + i2 * VEC) = yv2;
+    if (RELU && mask != nullptr) {
+      mask[i] = (unsigned char)mb;  // one byte per VEC-pack
+      if (has2) mask[i2] = (unsigned char)mb2;
+    }
   }
 }
 
@@ -435,6 +454,11 @@ std::vector<at::Tensor> batch_norm_fwd_train(
   auto scale = at::empty({C}, opts);
   auto shift = at::empty({C}, opts);
   auto y = at::empty_like(x);
+  // block-tail BNs (relu+addend) also emit a 1-bit relu mask (1/16 the
+  // bytes of y) so the bwd reduce skips the y read entirely
+  at::Tensor mask;
+  if (relu && addend)
+    mask = at::empty({R * C / 8}, x.options().dtype(at::kByte));
   auto stream = at::cuda::getCurrentCUDAStream();
 
   dispatch_vec(x, [&](auto* tp, auto vec) {
@@ -472,6 +496,7 @@ std::vector<at::Tensor> batch_norm_fwd_train(
   bn_apply_kernel<devT, VEC, RELU_, ADD_, true>                             \
       <<<agrid, AMD_TPB, smem, stream>>>(                                   \
           (const devT*)x.const_data_ptr(), zp, (devT*)y.data_ptr(),         \
+          mask.defined() ? mask.data_ptr<unsigned char>() : nullptr,        \
           scale.data_ptr<float>(), shift.data_ptr<float>(),     \
           total_vec, (int)C)
     if (relu && addend) APPLY(true, true);
@@ -481,7 +506,8 @@ std::vector<at::Tensor> batch_norm_fwd_train(
 #undef APPLY
     CHECK_CUDA_OK();
   });
-  return {y, mean, invstd, scale, shift};
+  if (!mask.defined()) mask = at::empty({0}, x.options().dtype(at::kByte));
+  return {y, mean, invstd, scale, shift, mask};
 }
 
 // forward-train using per-block statistics partials produced by the conv
@@ -501,6 +527,9 @@ std::vector<at::Tensor> batch_norm_fwd_train_from_parts(
   auto scale = at::empty({C}, opts);
   auto shift = at::empty({C}, opts);
   auto y = at::empty_like(x);
+  at::Tensor mask;
+  if (relu && addend)
+    mask = at::empty({R * C / 8}, x.options().dtype(at::kByte));
   auto stream = at::cuda::getCurrentCUDAStream();
 
   const int slots = bn_collapse_slots(nparts);
@@ -532,6 +561,7 @@ std::vector<at::Tensor> batch_norm_fwd_train_from_parts(
   bn_apply_kernel<devT, VEC, RELU_, ADD_, true>                             \
       <<<agrid, AMD_TPB, smem, stream>>>(                                   \
           (const devT*)x.const_data_ptr(), zp, (devT*)y.data_ptr(),         \
+          mask.defined() ? mask.data_ptr<unsigned char>() : nullptr,        \
           scale.data_ptr<float>(), shift.data_ptr<float>(),                 \
           total_vec, (int)C)
     if (relu && addend) APPLY2(true, true);
@@ -541,7 +571,8 @@ std::vector<at::Tensor> batch_norm_fwd_train_from_parts(
 #undef APPLY2
     CHECK_CUDA_OK();
   });
-  return {y, mean, invstd, scale, shift};
+  if (!mask.defined()) mask = at::empty({0}, x.options().dtype(at::kByte));
+  return {y, mean, invstd, scale, shift, mask};
 }
 
 at::Tensor batch_norm_fwd_eval(at::Tensor x, at::Tensor weight,
@@ -555,6 +586,7 @@ at::Tensor batch_norm_fwd_eval(at::Tensor x, at::Tensor weight,
   auto scale = at::empty({C}, opts);
   auto shift = at::empty({C}, opts);
   auto y = at::empty_like(x);
+  at::Tensor mask;  // eval: no mask output
   auto stream = at::cuda::getCurrentCUDAStream();
   int fgrid = (int)((C + AMD_TPB - 1) / AMD_TPB);
   bn_finalize_eval_kernel<<<fgrid, AMD_TPB, 0, stream>>>(
@@ -574,6 +606,7 @@ at::Tensor batch_norm_fwd_eval(at::Tensor x, at::Tensor weight,
   bn_apply_kernel<devT, VEC, RELU_, ADD_, true>                             \
       <<<agrid, AMD_TPB, smem, stream>>>(                                   \
           (const devT*)x.const_data_ptr(), zp, (devT*)y.data_ptr(),         \
+          mask.defined() ? mask.data_ptr<unsigned char>() : nullptr,        \
           scale.data_ptr<float>(), shift.data_ptr<float>(),     \
           total_vec, (int)C)
     if (relu && addend) APPLY(true, true);
@@ -592,7 +625,8 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
                                        bool relu, bool need_ghat,
                                        std::optional<at::Tensor> scale,
                                        std::optional<at::Tensor> shift,
-                                       std::optional<at::Tensor> grad_out2) {
+                                       std::optional<at::Tensor> grad_out2,
+                                       std::optional<at::Tensor> relu_mask) {
   // Pass structure (every kernel is an HBM-bound R x C stream, so passes
   // are the whole cost):
   //  * no relu:                reduce(x,go) -> apply(x,go -> gx)
@@ -640,12 +674,27 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
           invstd.data_ptr<float>(), scp, shp,                               \
           WG_ ? (devT*)ghat.data_ptr() : nullptr,                           \
           sums.data_ptr<float>(), R, (int)C)
-    if (mask_y && go2p) REDUCE(1, true, true);
+// MASK 3: the fwd bitmask rides in the y operand slot
+#define REDUCE3(GO2_)                                                       \
+  bn_reduce_kernel<devT, VEC, true, 3, true, GO2_>                          \
+      <<<rgrid, AMD_TPB, 0, stream>>>(                                      \
+          (const devT*)x.const_data_ptr(),                                  \
+          (const devT*)grad_out.const_data_ptr(), go2p,                     \
+          (const devT*)mkp, mean.data_ptr<float>(),                         \
+          invstd.data_ptr<float>(), scp, shp,                               \
+          (devT*)ghat.data_ptr(),                                           \
+          sums.data_ptr<float>(), R, (int)C)
+    const unsigned char* mkp = (relu_mask && relu_mask->numel())
+        ? relu_mask->data_ptr<unsigned char>() : nullptr;
+    if (mask_y && mkp && go2p) REDUCE3(true);
+    else if (mask_y && mkp) REDUCE3(false);
+    else if (mask_y && go2p) REDUCE(1, true, true);
     else if (mask_y) REDUCE(1, true, false);
     else if (affine_mask) REDUCE(2, false, false);
     else if (go2p) REDUCE(0, true, true);
     else REDUCE(0, false, false);
 #undef REDUCE
+#undef REDUCE3
     CHECK_CUDA_OK();
     const int slots = bn_collapse_slots(rgrid);
     auto sums2 = at::empty({slots, 2 * C}, opts);

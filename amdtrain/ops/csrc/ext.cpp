@@ -46,7 +46,8 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
                                        bool relu, bool need_ghat,
                                        std::optional<at::Tensor> scale,
                                        std::optional<at::Tensor> shift,
-                                       std::optional<at::Tensor> grad_out2);
+                                       std::optional<at::Tensor> grad_out2,
+                                       std::optional<at::Tensor> relu_mask);
 
 // gemm.hip
 at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out,
@@ -127,7 +128,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("y"), py::arg("weight"), py::arg("mean"), py::arg("invstd"),
         py::arg("relu"), py::arg("need_ghat"),
         py::arg("scale") = std::nullopt, py::arg("shift") = std::nullopt,
-        py::arg("grad_out2") = std::nullopt);
+        py::arg("grad_out2") = std::nullopt,
+        py::arg("relu_mask") = std::nullopt);
   m.def("gemm_bt", &gemm_bt, py::arg("A"), py::arg("B"),
         py::arg("f32_out") = false, py::arg("addend") = std::nullopt,
         py::arg("bias") = std::nullopt);

@@ -33,16 +33,19 @@ class _BNFunction(torch.autograd.Function):
                 momentum, eps, relu, addend, parts, res_cell=None):
         e = require_ext()
         if parts is not None:
-            y, mean, invstd, scale, shift = e.batch_norm_fwd_train_from_parts(
-                x, parts, weight, bias, running_mean, running_var,
-                float(momentum), float(eps), bool(relu), addend)
+            y, mean, invstd, scale, shift, mask = \
+                e.batch_norm_fwd_train_from_parts(
+                    x, parts, weight, bias, running_mean, running_var,
+                    float(momentum), float(eps), bool(relu), addend)
         else:
-            y, mean, invstd, scale, shift = e.batch_norm_fwd_train(
+            y, mean, invstd, scale, shift, mask = e.batch_norm_fwd_train(
                 x, weight, bias, running_mean, running_var,
                 float(momentum), float(eps), bool(relu), addend)
         # scale/shift ([C] fp32) let the bwd recompute the relu mask as
-        # scale*x+shift>0 for addend-free BNs — no y read in either bwd pass
-        ctx.save_for_backward(x, y, weight, mean, invstd, scale, shift)
+        # scale*x+shift>0 for addend-free BNs; block-tail BNs save the
+        # fwd-computed 1-bit relu mask instead (bwd skips the y read)
+        ctx.save_for_backward(x, y, weight, mean, invstd, scale, shift,
+                              mask)
         ctx.relu = bool(relu)
         ctx.has_addend = addend is not None
         # block-tail BN grad mailbox (see batch_norm below): our backward
@@ -54,7 +57,7 @@ class _BNFunction(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_out):
-        x, y, weight, mean, invstd, scale, shift = ctx.saved_tensors
+        x, y, weight, mean, invstd, scale, shift, mask = ctx.saved_tensors
         e = require_ext()
         go2 = None
         cell = getattr(ctx, "res_cell", None)
@@ -64,7 +67,7 @@ class _BNFunction(torch.autograd.Function):
         grad_x, grad_w, grad_b, ghat = e.batch_norm_bwd(
             x, grad_out.contiguous(memory_format=torch.channels_last),
             y, weight, mean, invstd, ctx.relu, ctx.has_addend,
-            scale, shift, go2)
+            scale, shift, go2, mask)
         grad_addend = ghat if ctx.has_addend else None
         return (grad_x, grad_w, grad_b, None, None, None, None, None,
                 grad_addend, None, None)
