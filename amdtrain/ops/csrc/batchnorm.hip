@@ -457,8 +457,8 @@ std::vector<at::Tensor> batch_norm_fwd_train(
   // block-tail BNs (relu+addend) also emit a 1-bit relu mask (1/16 the
   // bytes of y) so the bwd reduce skips the y read entirely
   at::Tensor mask;
-  if (relu && addend)
-    mask = at::empty({R * C / 8}, x.options().dtype(at::kByte));
+  if (relu && addend)  // one byte per VEC-pack; VEC is 4 for fp32 (16B/4B)
+    mask = at::empty({R * C / 4}, x.options().dtype(at::kByte));
   auto stream = at::cuda::getCurrentCUDAStream();
 
   dispatch_vec(x, [&](auto* tp, auto vec) {
@@ -528,8 +528,8 @@ std::vector<at::Tensor> batch_norm_fwd_train_from_parts(
   auto shift = at::empty({C}, opts);
   auto y = at::empty_like(x);
   at::Tensor mask;
-  if (relu && addend)
-    mask = at::empty({R * C / 8}, x.options().dtype(at::kByte));
+  if (relu && addend)  // one byte per VEC-pack; VEC is 4 for fp32 (16B/4B)
+    mask = at::empty({R * C / 4}, x.options().dtype(at::kByte));
   auto stream = at::cuda::getCurrentCUDAStream();
 
   const int slots = bn_collapse_slots(nparts);
