@@ -157,13 +157,20 @@ __device__ __forceinline__ void epilogue_stats(
   }
 }
 
-template <bool F32OUT, bool STRIDED>
+// NT: nontemporal C stores (global_store_* nt) — A/B experiment for the
+// ~4.2 TB/s streaming wall on the BW-bound shapes (AMDTRAIN_GEMM_NT=1)
+// NT=1: nontemporal C stores — measured 2x SLOWER (breaks write combining
+// of the 2 B scattered stores); kept env-gated as a documented negative.
+// LDSE=1: stage the C tile through LDS and emit coalesced 16 B stores
+// instead of per-element column-strided 2 B stores.
+template <bool F32OUT, bool STRIDED, bool NT = false, bool LDSE = false>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                void* __restrict__ C, long M, long N, long K, int nbm,
                int nbn, StrideMap sm, float* __restrict__ stats) {
-  __shared__ bf16 As[BM * BK];
-  __shared__ bf16 Bs[BN * BK];
+  __shared__ bf16 SMEM[(BM + BN) * BK];  // As | Bs (contiguous: the LDSE
+  bf16* const As = SMEM;                 // epilogue reuses all 16 KB)
+  bf16* const Bs = SMEM + BM * BK;
 
   const int nwg = nbm * nbn;
   const int bid = xcd_swizzle(blockIdx.x, nwg);
@@ -220,19 +227,53 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
   }
 
   // epilogue: C[m0+wm+i*16+fq*4+r][n0+wn+j*16+fr]
+  if (LDSE && !F32OUT && n0 + 128 <= N) {
+    // stage the bf16 C tile through LDS (reusing the 16 KB staging buffer,
+    // 64 rows per round) and emit coalesced 16 B row-major stores instead
+    // of the fragment-shaped column-strided 2 B stores
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
+    for (int h = 0; h < 2; ++h) {
+      __syncthreads();
+      if (wm == h * 64) {
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+        for (int i = 0; i < 4; ++i)
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        long row = m0 + wm + i * 16 + fq * 4 + r;
-        long col = n0 + wn + j * 16 + fr;
-        if (row < M && col < N) {
-          if (F32OUT)
-            ((float*)C)[row * N + col] = acc[i][j][r];
-          else
-            ((bf16*)C)[row * N + col] = __float2bfloat16(acc[i][j][r]);
+          for (int j = 0; j < 4; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+              SMEM[(i * 16 + fq * 4 + r) * 128 + wn + j * 16 + fr] =
+                  __float2bfloat16(acc[i][j][r]);
+      }
+      __syncthreads();
+      for (int u2 = t; u2 < 1024; u2 += GEMM_TPB) {
+        const int row_l = u2 >> 3;
+        const long row = m0 + h * 64 + row_l;
+        if (row < M)
+          *(uint4*)((bf16*)C + row * N + n0 + (u2 & 7) * 16) =
+              ((const uint4*)SMEM)[u2];
+      }
+    }
+  } else {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          long row = m0 + wm + i * 16 + fq * 4 + r;
+          long col = n0 + wn + j * 16 + fr;
+          if (row < M && col < N) {
+            if (F32OUT) {
+              ((float*)C)[row * N + col] = acc[i][j][r];
+            } else if (NT) {
+              bf16 v = __float2bfloat16(acc[i][j][r]);
+              short b;
+              __builtin_memcpy(&b, &v, 2);
+              __builtin_nontemporal_store(b, (short*)C + row * N + col);
+            } else {
+              ((bf16*)C)[row * N + col] = __float2bfloat16(acc[i][j][r]);
+            }
+          }
         }
       }
     }
@@ -642,8 +683,26 @@ at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out,
     CHECK_CUDA_OK();
     return C;
   }
+  static const bool use_nt = []() {
+    const char* v = std::getenv("AMDTRAIN_GEMM_NT");
+    return v && v[0] == '1';
+  }();
+  static const bool use_ldse = []() {
+    const char* v = std::getenv("AMDTRAIN_GEMM_LDSE");
+    return v && v[0] == '1';
+  }();
   if (f32_out)
     gemm_bt_kernel<true, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+        (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
+        C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr);
+  else if (use_ldse && N % 8 == 0)
+    gemm_bt_kernel<false, false, false, true>
+        <<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+            (const bf16*)Ac.const_data_ptr(),
+            (const bf16*)Bc.const_data_ptr(), C.data_ptr(), M, N, K, nbm,
+            nbn, sm, nullptr);
+  else if (use_nt)
+    gemm_bt_kernel<false, false, true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
         C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr);
   else
