@@ -73,3 +73,24 @@ def test_nccl_ddp_2proc(tmp_path):
                        capture_output=True, text=True, timeout=600)
     assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
     assert "NCCL_DDP_OK" in r.stdout
+
+
+def test_nccl_horovod_style_2proc(tmp_path):
+    """2-process horovod-style DistributedOptimizer over REAL RCCL with
+    fp16 gradient compression + rank-0 broadcasts (reference
+    horovod_distributed.py:149-164); asserts replica lockstep from
+    deliberately different inits."""
+    import os
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [sys.executable, "-m", "torch.distributed.run",
+           "--nnodes=1", "--nproc-per-node=2",
+           "--master-addr", "127.0.0.1", "--master-port", "29653",
+           os.path.join(root, "tests", "hvd_style_worker.py")]
+    env = dict(os.environ, PYTHONPATH=root + os.pathsep +
+               os.environ.get("PYTHONPATH", ""))
+    r = subprocess.run(cmd, cwd=str(tmp_path), env=env,
+                       capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "HVD_OK" in r.stdout
