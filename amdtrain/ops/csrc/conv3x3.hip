@@ -111,18 +111,21 @@ __device__ __forceinline__ long unit_row(const UnitCoord& u, int kh, int kw,
   return u.n_off + (long)h * g.W + w;
 }
 
-template <bool DGRAD>
+// BKT = K-step depth (32 or 64): BKT/8 16-B units per m-row, BKT/16 rounds
+template <bool DGRAD, int BKT>
 __device__ __forceinline__ void stage_gathered(
     const bf16* __restrict__ src, int ld, const UnitCoord* uc, int c0, int kh,
     int kw, const ConvGeom& g, const bf16* __restrict__ zero_page,
     bf16* lds) {
   const int t = threadIdx.x;
+  constexpr int UPR = BKT / 8;
 #pragma unroll
-  for (int rnd = 0; rnd < 2; ++rnd) {
-    int unit = rnd * GEMM_TPB + t;   // 0..511; 4 x 16B units per row
+  for (int rnd = 0; rnd < BKT / 16; ++rnd) {
+    int unit = rnd * GEMM_TPB + t;
     long row = unit_row<DGRAD>(uc[rnd], kh, kw, g);
     const bf16* p = row < 0 ? zero_page
-                            : src + row * (long)ld + c0 + (unit & 3) * 8;
+                            : src + row * (long)ld + c0 +
+                                  (unit % UPR) * 8;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)p,
         (__attribute__((address_space(3))) unsigned int*)(lds + unit * 8), 16,
@@ -132,16 +135,18 @@ __device__ __forceinline__ void stage_gathered(
 
 // stage a [128 rows][32 cols] tile of a plain [Rows x ld] matrix at column
 // offset koff (weights)
+template <int BKT>
 __device__ __forceinline__ void stage_plain(
     const bf16* __restrict__ gsrc, long ld, long row0, long rows, long koff,
     bf16* lds) {
   const int t = threadIdx.x;
+  constexpr int UPR = BKT / 8;
 #pragma unroll
-  for (int rnd = 0; rnd < 2; ++rnd) {
+  for (int rnd = 0; rnd < BKT / 16; ++rnd) {
     int unit = rnd * GEMM_TPB + t;
-    long row = row0 + (unit >> 2);
+    long row = row0 + unit / UPR;
     if (row >= rows) row = rows - 1;
-    const bf16* src = gsrc + row * ld + koff + (unit & 3) * 8;
+    const bf16* src = gsrc + row * ld + koff + (unit % UPR) * 8;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)src,
         (__attribute__((address_space(3))) unsigned int*)(lds + unit * 8), 16,
@@ -202,14 +207,14 @@ __device__ __forceinline__ void conv_epilogue_stats(
 // fwd / dgrad main kernel.  DGRAD only changes the gather map; operand roles:
 //   fwd:   A = x rows (AC channels), B = w [NC, 9*AC], C = y [M, NC]
 //   dgrad: A = dy rows (AC = Cout), B = w' [NC = Cin, 9*Cout], C = dx
-template <bool DGRAD>
+template <bool DGRAD, int BKT = BK>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
                bf16* __restrict__ C, long M, int AC, int NC, ConvGeom g,
                int nbm, int nbn, const bf16* __restrict__ zero_page,
                float* __restrict__ stats) {
-  __shared__ bf16 As[128 * BK];
-  __shared__ bf16 Bs[128 * BK];
+  __shared__ bf16 As[128 * BKT];
+  __shared__ bf16 Bs[128 * BKT];
 
   const int bid = xcd_swz(blockIdx.x, nbm * nbn);
   const int bm = bid / nbn, bn = bid % nbn;
@@ -227,37 +232,42 @@ conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   // hoist the per-unit coordinate decode out of the K-loop
-  UnitCoord uc[2];
+  UnitCoord uc[BKT / 16];
 #pragma unroll
-  for (int rnd = 0; rnd < 2; ++rnd) {
-    long m = m0 + ((rnd * GEMM_TPB + t) >> 2);
+  for (int rnd = 0; rnd < BKT / 16; ++rnd) {
+    long m = m0 + (rnd * GEMM_TPB + t) / (BKT / 8);
     if (m >= M) m = M - 1;
     uc[rnd] = decode_unit<DGRAD>(m, g);
   }
 
-  const int ksteps_per_tap = AC / BK;
+  const int ksteps_per_tap = AC / BKT;
   for (int tap = 0; tap < 9; ++tap) {
     const int kh = tap / 3, kw = tap % 3;
     for (int ks = 0; ks < ksteps_per_tap; ++ks) {
-      const int c0 = ks * BK;
+      const int c0 = ks * BKT;
       __syncthreads();
-      stage_gathered<DGRAD>(A, AC, uc, c0, kh, kw, g, zero_page, As);
-      stage_plain(Bw, (long)9 * AC, n0, NC, (long)tap * AC + c0, Bs);
+      stage_gathered<DGRAD, BKT>(A, AC, uc, c0, kh, kw, g, zero_page, As);
+      stage_plain<BKT>(Bw, (long)9 * AC, n0, NC, (long)tap * AC + c0, Bs);
       __syncthreads();
 
-      bf16x8 a[4], b[4];
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
-        a[i] = *(const bf16x8*)&As[(wm + i * 16 + fr) * BK + fq * 8];
+      for (int kk = 0; kk < BKT / 32; ++kk) {
+        bf16x8 a[4], b[4];
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
-        b[j] = *(const bf16x8*)&Bs[(wn + j * 16 + fr) * BK + fq * 8];
-#pragma unroll
-      for (int i = 0; i < 4; ++i)
+        for (int i = 0; i < 4; ++i)
+          a[i] = *(const bf16x8*)
+              &As[(wm + i * 16 + fr) * BKT + kk * 32 + fq * 8];
 #pragma unroll
         for (int j = 0; j < 4; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a[i], b[j], acc[i][j], 0, 0, 0);
+          b[j] = *(const bf16x8*)
+              &Bs[(wn + j * 16 + fr) * BKT + kk * 32 + fq * 8];
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a[i], b[j], acc[i][j], 0, 0, 0);
+      }
     }
   }
 
@@ -428,10 +438,22 @@ std::vector<at::Tensor> conv3x3_fwd_stats_impl(at::Tensor x2d, long Nn,
     stats_ptr = stats.data_ptr<float>();
   }
   auto stream = at::cuda::getCurrentCUDAStream();
-  conv3x3_kernel<false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
-      (const bf16*)x2d.const_data_ptr(), (const bf16*)w2d.const_data_ptr(),
-      (bf16*)y.data_ptr(), M, (int)Cin, (int)Cout, g, nbm, nbn,
-      (const bf16*)zp.const_data_ptr(), stats_ptr);
+  static const bool bk64 = []() {
+    const char* v = std::getenv("AMDTRAIN_CONV3X3_BK64");
+    return !(v && v[0] == '0');
+  }();
+  // A/B (tools/bench_conv3x3.py): BK64 fwd +17% at layer2-scale M, -4% at
+  // layer3/4 M; dgrad flat-to-negative -> fwd-only, large-M-only
+  if (bk64 && Cin % 64 == 0 && M >= 200000)
+    conv3x3_kernel<false, 64><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+        (const bf16*)x2d.const_data_ptr(), (const bf16*)w2d.const_data_ptr(),
+        (bf16*)y.data_ptr(), M, (int)Cin, (int)Cout, g, nbm, nbn,
+        (const bf16*)zp.const_data_ptr(), stats_ptr);
+  else
+    conv3x3_kernel<false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+        (const bf16*)x2d.const_data_ptr(), (const bf16*)w2d.const_data_ptr(),
+        (bf16*)y.data_ptr(), M, (int)Cin, (int)Cout, g, nbm, nbn,
+        (const bf16*)zp.const_data_ptr(), stats_ptr);
   CHECK_CUDA_OK();
   if (want_stats) return {y, stats};
   return {y};
@@ -467,10 +489,22 @@ at::Tensor conv3x3_dgrad(at::Tensor dy2d, long Nn, long H, long W,
   ConvGeom g{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
   int nbm = (int)((M + 127) / 128), nbn = (int)((Cin + 127) / 128);
   auto zp = zero_page_for(dy2d);
-  conv3x3_kernel<true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
-      (const bf16*)dy2d.const_data_ptr(), (const bf16*)wrot.const_data_ptr(),
-      (bf16*)dx.data_ptr(), M, (int)Cout, (int)Cin, g, nbm, nbn,
-      (const bf16*)zp.const_data_ptr(), nullptr);
+  static const bool bk64d = []() {  // measured negative for dgrad: off
+    const char* v = std::getenv("AMDTRAIN_CONV3X3_BK64D");
+    return v && v[0] == '1';
+  }();
+  if (bk64d && Cout % 64 == 0)
+    conv3x3_kernel<true, 64><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+        (const bf16*)dy2d.const_data_ptr(),
+        (const bf16*)wrot.const_data_ptr(), (bf16*)dx.data_ptr(), M,
+        (int)Cout, (int)Cin, g, nbm, nbn,
+        (const bf16*)zp.const_data_ptr(), nullptr);
+  else
+    conv3x3_kernel<true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+        (const bf16*)dy2d.const_data_ptr(),
+        (const bf16*)wrot.const_data_ptr(), (bf16*)dx.data_ptr(), M,
+        (int)Cout, (int)Cin, g, nbm, nbn,
+        (const bf16*)zp.const_data_ptr(), nullptr);
   CHECK_CUDA_OK();
   return dx;
 }
