@@ -249,3 +249,27 @@ def test_imagefolder_prefetcher_end_to_end(tmp_path):
         nb += 1
     assert nb == 4  # 16 train images / batch 4
     torch.cuda.synchronize()
+
+
+def test_cli_entrypoint_full_epoch_gpu(tmp_path):
+    """The real launcher-style entrypoint end-to-end ON GPU (world 1):
+    synthetic epoch -> train loop (custom kernels) -> distributed-eval
+    validate -> reference-schema checkpoint."""
+    import os
+    from amdtrain.cli.distributed import main
+
+    cwd = os.getcwd()
+    os.chdir(tmp_path)
+    try:
+        acc = main(["-a", "resnet50", "--synthetic",
+                    "--synthetic-train-size", "24",
+                    "--synthetic-val-size", "16",
+                    "--image-size", "64", "-b", "8", "--epochs", "1",
+                    "-j", "0", "-p", "1", "--dtype", "bf16"])
+    finally:
+        os.chdir(cwd)
+    assert isinstance(acc, float)
+    ck = torch.load(str(tmp_path / "checkpoint.pth.tar"),
+                    weights_only=True)
+    assert set(ck.keys()) == {"epoch", "arch", "state_dict", "best_acc1"}
+    assert ck["arch"] == "resnet50"
