@@ -44,6 +44,11 @@ def parse_args():
     p.add_argument("--style", type=str, default="ddp",
                    choices=["ddp", "apex", "horovod"],
                    help="launch-style variant to benchmark (BASELINE configs)")
+    p.add_argument("--hipgraph", type=str, default="auto",
+                   choices=["auto", "on", "off"],
+                   help="capture the steady-state step into a hipGraph and "
+                        "replay it (auto: on for world==1 ddp style — RCCL "
+                        "collectives stay outside capture)")
     return p.parse_args()
 
 
@@ -139,8 +144,28 @@ def main():
         if device.type == "cuda":
             torch.cuda.synchronize()
 
+    # hipGraph capture (north-star: HIP streams and graphs): the whole
+    # steady-state step — normalize, forward, backward, fused SGD — records
+    # once and replays with zero per-kernel launch overhead.  world>1 keeps
+    # eager launches (RCCL collectives inside capture are not exercised).
+    use_graph = (args.hipgraph == "on"
+                 or (args.hipgraph == "auto" and world == 1
+                     and args.style == "ddp" and device.type == "cuda"
+                     and not args.profile))
+    graph = None
+    if use_graph:
+        for _ in range(3):  # allocator warmup before capture
+            step()
+        sync()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            graph_loss = step()
+
+        def step_graph():
+            graph.replay()
+            return graph_loss
     for _ in range(args.warmup):
-        step()
+        (step_graph if graph is not None else step)()
     if args.profile and rank == 0:
         from amdtrain.utils.profiling import profile_steps
         with profile_steps(args.profile):
@@ -151,7 +176,7 @@ def main():
     sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        loss = step()
+        loss = (step_graph if graph is not None else step)()
     comm.barrier()
     sync()
     elapsed = time.perf_counter() - t0
@@ -188,6 +213,7 @@ def main():
                 "style": args.style,
                 "sec_per_epoch_est": round(1_281_167 / ips, 1),
                 "hip_ext": not args.no_ext,
+                "hipgraph": bool(graph is not None),
                 "peak_mem_gb": round(
                     torch.cuda.max_memory_allocated() / 2**30, 2)
                 if device.type == "cuda" else None,
