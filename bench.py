@@ -44,11 +44,13 @@ def parse_args():
     p.add_argument("--style", type=str, default="ddp",
                    choices=["ddp", "apex", "horovod"],
                    help="launch-style variant to benchmark (BASELINE configs)")
-    p.add_argument("--hipgraph", type=str, default="auto",
+    p.add_argument("--hipgraph", type=str, default="off",
                    choices=["auto", "on", "off"],
                    help="capture the steady-state step into a hipGraph and "
-                        "replay it (auto: on for world==1 ddp style — RCCL "
-                        "collectives stay outside capture)")
+                        "replay it (measured NEUTRAL at the default batch — "
+                        "the 313 ms step already hides launch overhead — so "
+                        "off by default; capture verified correct on "
+                        "hardware, world==1 only)")
     return p.parse_args()
 
 
