@@ -218,6 +218,8 @@ __global__ void bn_collapse_partials_kernel(const float* __restrict__ in,
 }
 
 // ---- finalize (train): stats + running update + scale/shift ---------------
+// 4 slot-threads per channel (the single-thread row loop over up to 128
+// collapse slots was ~20 us latency-bound x 106 calls/step)
 __global__ void bn_finalize_train_kernel(
     const float* __restrict__ partials, int nparts,
     const float* __restrict__ w,
@@ -225,13 +227,23 @@ __global__ void bn_finalize_train_kernel(
     float* __restrict__ rv, float* __restrict__ mean,
     float* __restrict__ invstd, float* __restrict__ scale,
     float* __restrict__ shift, long R, int C, float momentum, float eps) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int sl = threadIdx.x >> 6;  // 0..3
   float s1 = 0.f, s2 = 0.f;
-  for (int p = 0; p < nparts; ++p) {
-    s1 += partials[(long)p * 2 * C + c];
-    s2 += partials[(long)p * 2 * C + C + c];
-  }
+  if (c < C)
+    for (int p = sl; p < nparts; p += 4) {
+      s1 += partials[(long)p * 2 * C + c];
+      s2 += partials[(long)p * 2 * C + C + c];
+    }
+  __shared__ float red[2][4][64];
+  red[0][sl][threadIdx.x & 63] = s1;
+  red[1][sl][threadIdx.x & 63] = s2;
+  __syncthreads();
+  if (sl != 0 || c >= C) return;
+  s1 += red[0][1][threadIdx.x] + red[0][2][threadIdx.x] +
+        red[0][3][threadIdx.x];
+  s2 += red[1][1][threadIdx.x] + red[1][2][threadIdx.x] +
+        red[1][3][threadIdx.x];
   float m = s1 / R;
   float var = fmaxf(s2 / R - m * m, 0.f);
   float is = rsqrtf(var + eps);
@@ -324,13 +336,23 @@ __global__ void bn_bwd_finalize_kernel(
     const float* __restrict__ invstd, float* __restrict__ gw,
     float* __restrict__ gb, float* __restrict__ A, float* __restrict__ Bc,
     float* __restrict__ Dc, long R, int C) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int sl = threadIdx.x >> 6;
   float sg = 0.f, sgx = 0.f;
-  for (int p = 0; p < nparts; ++p) {
-    sg += partials[(long)p * 2 * C + c];
-    sgx += partials[(long)p * 2 * C + C + c];
-  }
+  if (c < C)
+    for (int p = sl; p < nparts; p += 4) {
+      sg += partials[(long)p * 2 * C + c];
+      sgx += partials[(long)p * 2 * C + C + c];
+    }
+  __shared__ float red[2][4][64];
+  red[0][sl][threadIdx.x & 63] = sg;
+  red[1][sl][threadIdx.x & 63] = sgx;
+  __syncthreads();
+  if (sl != 0 || c >= C) return;
+  sg += red[0][1][threadIdx.x] + red[0][2][threadIdx.x] +
+        red[0][3][threadIdx.x];
+  sgx += red[1][1][threadIdx.x] + red[1][2][threadIdx.x] +
+         red[1][3][threadIdx.x];
   gb[c] = sg;
   gw[c] = sgx;
   float a = w[c] * invstd[c];
@@ -479,7 +501,7 @@ std::vector<at::Tensor> batch_norm_fwd_train(
           (int)(2 * C), slots);
       CHECK_CUDA_OK();
     }
-    int fgrid = (int)((C + AMD_TPB - 1) / AMD_TPB);
+    int fgrid = (int)((C + 63) / 64);
     bn_finalize_train_kernel<<<fgrid, AMD_TPB, 0, stream>>>(
         sums2.data_ptr<float>(), slots, weight.data_ptr<float>(),
         bias.data_ptr<float>(), running_mean.data_ptr<float>(),
@@ -541,7 +563,7 @@ std::vector<at::Tensor> batch_norm_fwd_train_from_parts(
         (int)(2 * C), slots);
     CHECK_CUDA_OK();
   }
-  int fgrid = (int)((C + AMD_TPB - 1) / AMD_TPB);
+  int fgrid = (int)((C + 63) / 64);
   bn_finalize_train_kernel<<<fgrid, AMD_TPB, 0, stream>>>(
       sums2.data_ptr<float>(), slots, weight.data_ptr<float>(),
       bias.data_ptr<float>(), running_mean.data_ptr<float>(),
@@ -705,7 +727,7 @@ std::vector<at::Tensor> batch_norm_bwd(at::Tensor x, at::Tensor grad_out,
           (int)(2 * C), slots);
       CHECK_CUDA_OK();
     }
-    int fgrid = (int)((C + AMD_TPB - 1) / AMD_TPB);
+    int fgrid = (int)((C + 63) / 64);
     bn_bwd_finalize_kernel<<<fgrid, AMD_TPB, 0, stream>>>(
         sums2.data_ptr<float>(), slots, weight.data_ptr<float>(),
         mean.data_ptr<float>(), invstd.data_ptr<float>(),
