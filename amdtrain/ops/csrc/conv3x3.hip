@@ -442,9 +442,10 @@ std::vector<at::Tensor> conv3x3_fwd_stats_impl(at::Tensor x2d, long Nn,
     const char* v = std::getenv("AMDTRAIN_CONV3X3_BK64");
     return !(v && v[0] == '0');
   }();
-  // A/B (tools/bench_conv3x3.py): BK64 fwd +17% at layer2-scale M, -4% at
-  // layer3/4 M; dgrad flat-to-negative -> fwd-only, large-M-only
-  if (bk64 && Cin % 64 == 0 && M >= 200000)
+  // A/B (tools/bench_conv3x3.py + in-context traces): BK64 fwd +17% at
+  // layer2 (Cin=128, M=400k); layer1 (Cin=64) and layer3/4 (small M)
+  // measured WORSE in-context -> Cin>=128 && large-M only
+  if (bk64 && Cin % 64 == 0 && Cin >= 128 && M >= 200000)
     conv3x3_kernel<false, 64><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)x2d.const_data_ptr(), (const bf16*)w2d.const_data_ptr(),
         (bf16*)y.data_ptr(), M, (int)Cin, (int)Cout, g, nbm, nbn,
