@@ -59,3 +59,36 @@ def test_pool_cpu_paths():
                           F.max_pool2d(x, 3, 2, 1))
     assert torch.allclose(OF.global_avg_pool(x),
                           F.adaptive_avg_pool2d(x, (1, 1)))
+
+
+def test_residual_tap_unarmed_passthrough():
+    """ResidualGradTap must behave as identity for autograd unless the
+    producing BN armed the mailbox (the safety latch that keeps every
+    fallback path correct)."""
+    import torch
+    from amdtrain.ops.conv import GradCell, ResidualGradTap
+
+    cell = GradCell()
+    x = torch.randn(4, requires_grad=True)
+    (ResidualGradTap.apply(x, cell) * 2).sum().backward()
+    assert torch.allclose(x.grad, torch.full((4,), 2.0))
+    assert cell.g is None
+
+    cell2 = GradCell()
+    cell2.armed = True
+    x2 = torch.randn(4, requires_grad=True)
+    ResidualGradTap.apply(x2, cell2).sum().backward()
+    assert x2.grad is None  # rerouted, not accumulated
+    assert torch.allclose(cell2.g, torch.ones(4))
+
+
+def test_gradcell_survives_amp_cast():
+    """GradCell instances must pass through torch.amp.custom_fwd's
+    cast_inputs traversal UNTOUCHED (a dict would be deep-copied and
+    silently disconnect the mailbox — the round-2 bug this guards)."""
+    import torch
+    from amdtrain.ops.conv import GradCell
+
+    cell = GradCell()
+    out = torch.amp.autocast_mode._cast((cell,), "cuda", torch.bfloat16)
+    assert out[0] is cell
