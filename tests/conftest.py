@@ -17,6 +17,11 @@ def pytest_configure(config):
 
 def pytest_collection_modifyitems(config, items):
     import torch
+    # global hang insurance (pytest-timeout): no single test may eat the
+    # CI/driver budget; multi-process launch tests get headroom
+    for item in items:
+        if item.get_closest_marker("timeout") is None:
+            item.add_marker(pytest.mark.timeout(600))
     if torch.cuda.is_available():
         return
     skip_gpu = pytest.mark.skip(reason="no GPU in this environment")
