@@ -412,8 +412,11 @@ at::Tensor tn2_wgrad(at::Tensor dY, at::Tensor X, long taps, long Nn, long H,
   // config: (2,2) 128x128 tile MC=64 default; skinny-N with wide K -> (1,4);
   // skinny-K with tall N -> (4,1) (both MC=32 to keep 2 blocks/CU in LDS)
   if (N < 128 && K9 >= 256) {
+    // skinny-N conv wgrad: the (2,2,64) tile zero-pads the N half but its
+    // MC=64 pipeline beats the fully-dense (1,4,32) config (which is
+    // issue-bound at 16 MFMA/barrier); A/B via AMDTRAIN_TN2_14
     if (gmode == 2 && std::getenv("AMDTRAIN_TN2_14") == nullptr)
-      tn2_launch<2, 1, 2, 64>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
+      tn2_launch<2, 2, 2, 64>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
     else if (gmode == 2)
       tn2_launch<2, 1, 4, 32>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
     else if (gmode == 1)
