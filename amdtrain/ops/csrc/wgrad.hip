@@ -149,6 +149,21 @@ __device__ __forceinline__ void stage_tn2(
   }
 }
 
+template <int N>
+__device__ __forceinline__ void vmcnt_wait() {
+  if constexpr (N == 0) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  else if constexpr (N == 2) asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+  else if constexpr (N == 4) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  else if constexpr (N == 5) asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+  else if constexpr (N == 6) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  else if constexpr (N == 8) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  else if constexpr (N == 10)
+    asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+  else if constexpr (N == 16)
+    asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
+  else static_assert(N == 0, "unsupported vmcnt");
+}
+
 // Two hardware transpose reads -> one bf16x8 MFMA operand.
 // addr = LDS tile base + lane*8 + (runtime wave column offset); OFF is the
 // compile-time byte offset of the (sub, r=0) group; r=1 is the next group.
@@ -173,7 +188,12 @@ __device__ __forceinline__ bf16x8 tr_frag(unsigned addr) {
 
 // NW x KW = wave grid over the (n, k) output tile (NW*KW == 4); per-wave
 // sub-tile is 64x64 (4x4 fragments).  MC = m rows staged per chunk.
-template <int GMODE, int NW, int KW, int MC>
+// NBUF = LDS pipeline depth: 2 = classic double buffer with a full
+// vmcnt(0) drain per chunk; 3 = stage TWO chunks ahead and only wait for
+// the NEWEST chunk's loads at each barrier (counted vmcnt, guide T4) —
+// each wave drains to <= its own newest-chunk loads before the barrier,
+// so the chunk consumed next iteration is complete for every wave.
+template <int GMODE, int NW, int KW, int MC, int NBUF = 2>
 __global__ void __launch_bounds__(TN2_TPB, 2)
 tn2_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
            float* __restrict__ parts, long M, int N, int K9, int Cin,
@@ -182,8 +202,11 @@ tn2_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
   constexpr int NCY = NW * 64;   // dY tile cols
   constexpr int NCX = KW * 64;   // X tile cols
   constexpr int KSTEPS = MC / 32;
-  __shared__ __align__(16) char Ys[2][tile_bytes<NCY, MC>()];
-  __shared__ __align__(16) char Xs[2][tile_bytes<NCX, MC>()];
+  // per-thread global_load_lds issues per chunk (both operands):
+  // subtiles = (MC/32)*(cols/16) per operand, one wave-round each -> /4
+  constexpr int LC = (MC / 32) * (NCY / 16 + NCX / 16) / 4;
+  __shared__ __align__(16) char Ys[NBUF][tile_bytes<NCY, MC>()];
+  __shared__ __align__(16) char Xs[NBUF][tile_bytes<NCX, MC>()];
 
   const int tiles = nbn * nbk;
   const int bid = xcd_swz_tn2(blockIdx.x, tiles * msplit);
@@ -223,15 +246,25 @@ tn2_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
     stage_tn2<0, NCY, MC>(dY, N, mc0 * MC, M, n0, N, Cin, geo, zp, (bf16*)Ys[0]);
     stage_tn2<GMODE, NCX, MC>(X, Cin, mc0 * MC, M, k0, K9, Cin, geo, zp,
                               (bf16*)Xs[0]);
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    if (NBUF >= 3 && mc0 + 1 < mc1) {
+      stage_tn2<0, NCY, MC>(dY, N, (mc0 + 1) * MC, M, n0, N, Cin, geo, zp,
+                            (bf16*)Ys[1]);
+      stage_tn2<GMODE, NCX, MC>(X, Cin, (mc0 + 1) * MC, M, k0, K9, Cin, geo,
+                                zp, (bf16*)Xs[1]);
+      vmcnt_wait<LC>();  // chunk 0 complete; chunk 1 may stay in flight
+    } else {
+      vmcnt_wait<0>();
+    }
     __builtin_amdgcn_s_barrier();
 
     for (long mc = mc0; mc < mc1; ++mc) {
-      const int cur = (int)(mc - mc0) & 1, nxt = cur ^ 1;
-      if (mc + 1 < mc1) {
-        stage_tn2<0, NCY, MC>(dY, N, (mc + 1) * MC, M, n0, N, Cin, geo, zp,
+      const int cur = (int)((mc - mc0) % NBUF);
+      const long pre = NBUF >= 3 ? mc + 2 : mc + 1;
+      if (pre < mc1) {
+        const int nxt = (int)((pre - mc0) % NBUF);
+        stage_tn2<0, NCY, MC>(dY, N, pre * MC, M, n0, N, Cin, geo, zp,
                               (bf16*)Ys[nxt]);
-        stage_tn2<GMODE, NCX, MC>(X, Cin, (mc + 1) * MC, M, k0, K9, Cin, geo,
+        stage_tn2<GMODE, NCX, MC>(X, Cin, pre * MC, M, k0, K9, Cin, geo,
                                   zp, (bf16*)Xs[nxt]);
       }
       const unsigned yb = ybase0 + cur * YB;
@@ -258,7 +291,10 @@ tn2_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
                 a[i], b[j], acc[i][j], 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
       }
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      if (NBUF >= 3 && mc + 2 < mc1)
+        vmcnt_wait<LC>();  // only the newest chunk's loads may remain
+      else
+        vmcnt_wait<0>();
       __builtin_amdgcn_s_barrier();
     }
   }
@@ -329,7 +365,7 @@ static at::Tensor tn2_zero_page(const at::Tensor& like) {
   return zp;
 }
 
-template <int GMODE, int NW, int KW, int MC>
+template <int GMODE, int NW, int KW, int MC, int NBUF = 2>
 void tn2_launch(const at::Tensor& dY, const at::Tensor& X, at::Tensor& out,
                 long M, int N, int K9, int Cin, TnGeom geo, int target_blocks,
                 hipStream_t stream) {
@@ -343,7 +379,7 @@ void tn2_launch(const at::Tensor& dY, const at::Tensor& X, at::Tensor& out,
   at::Tensor parts = out;
   if (msplit > 1)
     parts = at::empty({msplit, (long)N * K9}, out.options());
-  tn2_kernel<GMODE, NW, KW, MC>
+  tn2_kernel<GMODE, NW, KW, MC, NBUF>
       <<<(int)(tiles * msplit), TN2_TPB, 0, stream>>>(
           (const bf16*)dY.const_data_ptr(), (const bf16*)X.const_data_ptr(),
           parts.data_ptr<float>(), M, N, K9, Cin, geo, nbn, nbk, msplit,
@@ -426,12 +462,30 @@ at::Tensor tn2_wgrad(at::Tensor dY, at::Tensor X, long taps, long Nn, long H,
   } else if (K9 < 128 && N >= 256 && gmode == 0) {
     tn2_launch<0, 4, 1, 32>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
   } else {
-    if (gmode == 2)
+    static const bool pipe3 = []() {
+      const char* v = std::getenv("AMDTRAIN_TN2_PIPE3");
+      return v && v[0] == '1';
+    }();
+    if (pipe3) {
+      // A/B: 3-buffer counted-vmcnt pipeline at MC=32 (48 KB LDS -> 3
+      // blocks/CU).  MEASURED -10..-15% vs the MC=64 double buffer
+      // (docs/KERNELS.md negative-results) — kept off-by-default
+      if (gmode == 2)
+        tn2_launch<2, 2, 2, 32, 3>(Yc, Xc, out, M, N, K9, Cin, geo, target,
+                                   stream);
+      else if (gmode == 1)
+        tn2_launch<1, 2, 2, 32, 3>(Yc, Xc, out, M, N, K9, Cin, geo, target,
+                                   stream);
+      else
+        tn2_launch<0, 2, 2, 32, 3>(Yc, Xc, out, M, N, K9, Cin, geo, target,
+                                   stream);
+    } else if (gmode == 2) {
       tn2_launch<2, 2, 2, 64>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
-    else if (gmode == 1)
+    } else if (gmode == 1) {
       tn2_launch<1, 2, 2, 64>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
-    else
+    } else {
       tn2_launch<0, 2, 2, 64>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
+    }
   }
   return out;
 }
