@@ -204,6 +204,84 @@ def conv3x3_mfma(x: torch.Tensor, weight: torch.Tensor,
     return y
 
 
+class _ConvGrouped3x3(torch.autograd.Function):
+    """Grouped 3x3 conv (ResNeXt) as a BLOCK-DIAGONALIZED dense conv over
+    the in-house implicit-GEMM kernels.
+
+    The grouped weight [Cout, S, 3, 3] (S = Cin/groups) is expanded to a
+    dense [Cout, 9*Cin] with zeros off the group diagonal, then the
+    standard conv3x3 fwd/dgrad/wgrad kernels run at dense speed (~640 TF)
+    — MIOpen's grouped path measured ~4x slower end-to-end on
+    resnext50_32x4d.  Numerically exact: zeros cannot contaminate fwd or
+    dgrad, and wgrad is an outer product, so slicing the group diagonal
+    of the dense dW recovers the grouped gradient bit-for-bit.
+    """
+
+    @staticmethod
+    @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
+    def forward(ctx, x: torch.Tensor, weight: torch.Tensor, stride: int,
+                groups: int):
+        e = require_ext()
+        n, cin, h, w = x.shape
+        cout = weight.shape[0]
+        sg = cin // groups
+        xc = x.contiguous(memory_format=torch.channels_last)
+        x2d = _rows(xc)
+        # dense [G, Sout, 3, 3, G, Sin] with the grouped weight on the
+        # (g, :, :, :, g, :) diagonal
+        sout = cout // groups
+        wg = weight.contiguous(memory_format=torch.channels_last) \
+            .permute(0, 2, 3, 1).reshape(groups, sout, 3, 3, sg)
+        dense = torch.zeros(groups, sout, 3, 3, groups, sg,
+                            dtype=wg.dtype, device=wg.device)
+        gi = torch.arange(groups, device=wg.device)
+        dense[gi, :, :, :, gi, :] = wg
+        w2d = dense.reshape(cout, 3, 3, cin).reshape(cout, 9 * cin)
+        if _fuse_stats_enabled():
+            y2d, stats = e.conv3x3_fwd_stats(x2d, n, h, w, stride, w2d)
+        else:
+            y2d = e.conv3x3_fwd(x2d, n, h, w, stride, w2d)
+            stats = torch.empty(0, device=x.device)
+        ctx.save_for_backward(x2d, w2d)
+        ctx.meta = (n, cin, h, w, stride, cout, groups)
+        ho = (h + 2 - 3) // stride + 1
+        wo = (w + 2 - 3) // stride + 1
+        y = y2d.view(n, ho, wo, cout).permute(0, 3, 1, 2)
+        ctx.mark_non_differentiable(stats)
+        return y, stats
+
+    @staticmethod
+    @torch.amp.custom_bwd(device_type="cuda")
+    def backward(ctx, grad_y: torch.Tensor, _grad_stats=None):
+        e = require_ext()
+        x2d, w2d = ctx.saved_tensors
+        n, cin, h, w, stride, cout, groups = ctx.meta
+        sg = cin // groups
+        sout = cout // groups
+        gy = grad_y.contiguous(memory_format=torch.channels_last)
+        gy2d = _rows(gy).to(torch.bfloat16)
+        dx2d = e.conv3x3_dgrad(gy2d, n, h, w, stride, w2d)
+        dx = dx2d.view(n, h, w, cin).permute(0, 3, 1, 2)
+        dw2d = e.tn2_wgrad(gy2d, x2d, 9, n, h, w, stride, 2) \
+            if _wgrad2_enabled() \
+            else e.conv3x3_wgrad(gy2d, x2d, n, h, w, stride)
+        # slice the group diagonal of the dense dW
+        dwd = dw2d.view(groups, sout, 3, 3, groups, sg)
+        gi = torch.arange(groups, device=dwd.device)
+        dw = dwd[gi, :, :, :, gi, :].permute(0, 1, 4, 2, 3) \
+            .reshape(cout, sg, 3, 3) \
+            .contiguous(memory_format=torch.channels_last)
+        return dx, dw, None, None
+
+
+def conv3x3_grouped_mfma(x: torch.Tensor, weight: torch.Tensor,
+                         stride: int, groups: int) -> torch.Tensor:
+    y, stats = _ConvGrouped3x3.apply(x, weight, stride, groups)
+    if stats.numel():
+        y._amdtrain_bn_stats = stats
+    return y
+
+
 class _ConvGeneric(torch.autograd.Function):
     """Generic implicit-GEMM conv (element-gather im2col) — serves the 7x7
     stem and any other odd configuration.  Input gradient gathers dY
@@ -302,6 +380,20 @@ class AmdConv2d(nn.Conv2d):
         # fp32 inference/training falls through to MIOpen
         bf16_ok = (x.dtype == torch.bfloat16
                    or torch.is_autocast_enabled("cuda"))
+        if (x.is_cuda and bf16_ok and ext_available()
+                and self.bias is None and self.dilation == (1, 1)
+                and self.groups > 1):
+            # grouped 3x3 (ResNeXt): block-diagonalized dense path
+            if (self.kernel_size == (3, 3) and self.padding == (1, 1)
+                    and self.stride[0] in (1, 2)
+                    and self.in_channels % 32 == 0
+                    and self.out_channels % 16 == 0
+                    and self.in_channels % self.groups == 0
+                    and os.environ.get("AMDTRAIN_CONV3X3G", "custom")
+                    == "custom"):
+                return conv3x3_grouped_mfma(x, self.weight, self.stride[0],
+                                            self.groups)
+            return super().forward(x)
         if (x.is_cuda and bf16_ok and ext_available()
                 and self.bias is None and self.groups == 1
                 and self.dilation == (1, 1)):
