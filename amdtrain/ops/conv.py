@@ -237,13 +237,17 @@ class _ConvGrouped3x3(torch.autograd.Function):
         gi = torch.arange(groups, device=wg.device)
         dense[gi, :, :, :, gi, :] = wg
         w2d = dense.reshape(cout, 3, 3, cin).reshape(cout, 9 * cin)
+        # group-diagonal banding: skip the provably-zero K-steps (the
+        # 128-channel window of each n-tile) when alignment allows
+        banded = (cin == cout and cin % 128 == 0)
         if _fuse_stats_enabled():
-            y2d, stats = e.conv3x3_fwd_stats(x2d, n, h, w, stride, w2d)
+            y2d, stats = e.conv3x3_fwd_stats(x2d, n, h, w, stride, w2d,
+                                             banded)
         else:
             y2d = e.conv3x3_fwd(x2d, n, h, w, stride, w2d)
             stats = torch.empty(0, device=x.device)
         ctx.save_for_backward(x2d, w2d)
-        ctx.meta = (n, cin, h, w, stride, cout, groups)
+        ctx.meta = (n, cin, h, w, stride, cout, groups, banded)
         ho = (h + 2 - 3) // stride + 1
         wo = (w + 2 - 3) // stride + 1
         y = y2d.view(n, ho, wo, cout).permute(0, 3, 1, 2)
@@ -255,12 +259,12 @@ class _ConvGrouped3x3(torch.autograd.Function):
     def backward(ctx, grad_y: torch.Tensor, _grad_stats=None):
         e = require_ext()
         x2d, w2d = ctx.saved_tensors
-        n, cin, h, w, stride, cout, groups = ctx.meta
+        n, cin, h, w, stride, cout, groups, banded = ctx.meta
         sg = cin // groups
         sout = cout // groups
         gy = grad_y.contiguous(memory_format=torch.channels_last)
         gy2d = _rows(gy).to(torch.bfloat16)
-        dx2d = e.conv3x3_dgrad(gy2d, n, h, w, stride, w2d)
+        dx2d = e.conv3x3_dgrad(gy2d, n, h, w, stride, w2d, banded)
         dx = dx2d.view(n, h, w, cin).permute(0, 3, 1, 2)
         dw2d = e.tn2_wgrad(gy2d, x2d, 9, n, h, w, stride, 2) \
             if _wgrad2_enabled() \

@@ -207,7 +207,11 @@ __device__ __forceinline__ void conv_epilogue_stats(
 // fwd / dgrad main kernel.  DGRAD only changes the gather map; operand roles:
 //   fwd:   A = x rows (AC channels), B = w [NC, 9*AC], C = y [M, NC]
 //   dgrad: A = dy rows (AC = Cout), B = w' [NC = Cin, 9*Cout], C = dx
-template <bool DGRAD, int BKT = BK>
+// BAND: block-diagonal weight (grouped conv via the dense path, Cin==NC,
+// both % 128): only the K-steps whose 128-channel window matches this
+// n-tile's group window are nonzero — skip the rest (groups/128-per-S of
+// the work)
+template <bool DGRAD, int BKT = BK, bool BAND = false>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
                bf16* __restrict__ C, long M, int AC, int NC, ConvGeom g,
@@ -240,10 +244,11 @@ conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
     uc[rnd] = decode_unit<DGRAD>(m, g);
   }
 
-  const int ksteps_per_tap = AC / BKT;
+  const int ks_lo = BAND ? (int)(n0 / BKT) : 0;
+  const int ks_hi = BAND ? ks_lo + 128 / BKT : AC / BKT;
   for (int tap = 0; tap < 9; ++tap) {
     const int kh = tap / 3, kw = tap % 3;
-    for (int ks = 0; ks < ksteps_per_tap; ++ks) {
+    for (int ks = ks_lo; ks < ks_hi; ++ks) {
       const int c0 = ks * BKT;
       __syncthreads();
       stage_gathered<DGRAD, BKT>(A, AC, uc, c0, kh, kw, g, zero_page, As);
@@ -419,7 +424,8 @@ static at::Tensor zero_page_for(const at::Tensor& like) {
 std::vector<at::Tensor> conv3x3_fwd_stats_impl(at::Tensor x2d, long Nn,
                                                 long H, long W, long stride,
                                                 at::Tensor w2d,
-                                                bool want_stats) {
+                                                bool want_stats,
+                                                bool banded = false) {
   // x2d: [Nn*H*W, Cin] bf16 NHWC rows; w2d: [Cout, 9*Cin]
   TORCH_CHECK(x2d.is_cuda() && x2d.scalar_type() == at::kBFloat16);
   long Cin = x2d.size(1), Cout = w2d.size(0);
@@ -445,7 +451,12 @@ std::vector<at::Tensor> conv3x3_fwd_stats_impl(at::Tensor x2d, long Nn,
   // A/B (tools/bench_conv3x3.py + in-context traces): BK64 fwd +17% at
   // layer2 (Cin=128, M=400k); layer1 (Cin=64) and layer3/4 (small M)
   // measured WORSE in-context -> Cin>=128 && large-M only
-  if (bk64 && Cin % 64 == 0 && Cin >= 128 && M >= 200000)
+  if (banded)  // requires Cin == Cout, both % 128 (host-checked)
+    conv3x3_kernel<false, BK, true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+        (const bf16*)x2d.const_data_ptr(), (const bf16*)w2d.const_data_ptr(),
+        (bf16*)y.data_ptr(), M, (int)Cin, (int)Cout, g, nbm, nbn,
+        (const bf16*)zp.const_data_ptr(), stats_ptr);
+  else if (bk64 && Cin % 64 == 0 && Cin >= 128 && M >= 200000)
     conv3x3_kernel<false, 64><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)x2d.const_data_ptr(), (const bf16*)w2d.const_data_ptr(),
         (bf16*)y.data_ptr(), M, (int)Cin, (int)Cout, g, nbm, nbn,
@@ -467,12 +478,15 @@ at::Tensor conv3x3_fwd(at::Tensor x2d, long Nn, long H, long W, long stride,
 
 std::vector<at::Tensor> conv3x3_fwd_stats(at::Tensor x2d, long Nn, long H,
                                           long W, long stride,
-                                          at::Tensor w2d) {
-  return conv3x3_fwd_stats_impl(x2d, Nn, H, W, stride, w2d, true);
+                                          at::Tensor w2d, bool banded) {
+  TORCH_CHECK(!banded || (x2d.size(1) == w2d.size(0) &&
+                          x2d.size(1) % 128 == 0),
+              "banded conv3x3 needs Cin == Cout, both % 128");
+  return conv3x3_fwd_stats_impl(x2d, Nn, H, W, stride, w2d, true, banded);
 }
 
 at::Tensor conv3x3_dgrad(at::Tensor dy2d, long Nn, long H, long W,
-                         long stride, at::Tensor w2d) {
+                         long stride, at::Tensor w2d, bool banded) {
   // dy2d: [Nn*Hout*Wout, Cout]; returns dx2d [Nn*H*W, Cin]
   long Cout = dy2d.size(1), Cin = w2d.size(1) / 9;
   TORCH_CHECK(w2d.size(0) == Cout && w2d.size(1) == 9 * Cin);
@@ -490,6 +504,16 @@ at::Tensor conv3x3_dgrad(at::Tensor dy2d, long Nn, long H, long W,
   ConvGeom g{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
   int nbm = (int)((M + 127) / 128), nbn = (int)((Cin + 127) / 128);
   auto zp = zero_page_for(dy2d);
+  if (banded) {
+    TORCH_CHECK(Cin == Cout && Cin % 128 == 0);
+    conv3x3_kernel<true, BK, true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+        (const bf16*)dy2d.const_data_ptr(),
+        (const bf16*)wrot.const_data_ptr(), (bf16*)dx.data_ptr(), M,
+        (int)Cout, (int)Cin, g, nbm, nbn,
+        (const bf16*)zp.const_data_ptr(), nullptr);
+    CHECK_CUDA_OK();
+    return dx;
+  }
   static const bool bk64d = []() {  // measured negative for dgrad: off
     const char* v = std::getenv("AMDTRAIN_CONV3X3_BK64D");
     return v && v[0] == '1';

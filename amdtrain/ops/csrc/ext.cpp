@@ -65,7 +65,7 @@ at::Tensor conv3x3_fwd(at::Tensor x2d, long Nn, long H, long W, long stride,
                        at::Tensor w2d);
 std::vector<at::Tensor> conv3x3_fwd_stats(at::Tensor x2d, long Nn, long H,
                                           long W, long stride,
-                                          at::Tensor w2d);
+                                          at::Tensor w2d, bool banded);
 std::vector<at::Tensor> gemm_bt_stats(at::Tensor A, at::Tensor B);
 
 // gemm8p.hip
@@ -74,7 +74,7 @@ at::Tensor gemm_bt_8p3(at::Tensor A, at::Tensor B);
 at::Tensor conv3x3_8p(at::Tensor A2d, long Nn, long H, long W, long stride,
                       at::Tensor w2d, bool dgrad);
 at::Tensor conv3x3_dgrad(at::Tensor dy2d, long Nn, long H, long W,
-                         long stride, at::Tensor w2d);
+                         long stride, at::Tensor w2d, bool banded);
 at::Tensor conv3x3_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn, long H,
                          long W, long stride);
 
@@ -139,12 +139,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_tn_strided", &gemm_tn_strided);
   m.def("transpose_2d", &transpose_2d);
   m.def("conv3x3_fwd", &conv3x3_fwd);
-  m.def("conv3x3_fwd_stats", &conv3x3_fwd_stats);
+  m.def("conv3x3_fwd_stats", &conv3x3_fwd_stats, py::arg("x2d"),
+        py::arg("Nn"), py::arg("H"), py::arg("W"), py::arg("stride"),
+        py::arg("w2d"), py::arg("banded") = false);
   m.def("gemm_bt_stats", &gemm_bt_stats);
   m.def("gemm_bt_8p", &gemm_bt_8p);
   m.def("gemm_bt_8p3", &gemm_bt_8p3);
   m.def("conv3x3_8p", &conv3x3_8p);
-  m.def("conv3x3_dgrad", &conv3x3_dgrad);
+  m.def("conv3x3_dgrad", &conv3x3_dgrad, py::arg("dy2d"), py::arg("Nn"),
+        py::arg("H"), py::arg("W"), py::arg("stride"), py::arg("w2d"),
+        py::arg("banded") = false);
   m.def("conv3x3_wgrad", &conv3x3_wgrad);
   m.def("tn2_wgrad", &tn2_wgrad, py::arg("dY"), py::arg("X"),
         py::arg("taps") = 1, py::arg("Nn") = 0, py::arg("H") = 0,
