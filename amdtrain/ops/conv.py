@@ -266,6 +266,17 @@ class _ConvGrouped3x3(torch.autograd.Function):
         gy2d = _rows(gy).to(torch.bfloat16)
         dx2d = e.conv3x3_dgrad(gy2d, n, h, w, stride, w2d, banded)
         dx = dx2d.view(n, h, w, cin).permute(0, 3, 1, 2)
+        if banded and _wgrad2_enabled():
+            # compact band [Cout, 9, 128]: extract the group diagonal of
+            # each 128-channel window
+            dwb = e.tn2_wgrad_banded(gy2d, x2d, n, h, w, stride) \
+                .view(cout // 128, 128 // sg, sg, 9, 128 // sg, sg)
+            bi = torch.arange(cout // 128, device=dwb.device)[:, None]
+            gi = torch.arange(128 // sg, device=dwb.device)[None, :]
+            dw = dwb[bi, gi, :, :, gi, :].reshape(cout, 9, sg) \
+                .view(cout, 3, 3, sg).permute(0, 3, 1, 2) \
+                .contiguous(memory_format=torch.channels_last)
+            return dx, dw, None, None
         dw2d = e.tn2_wgrad(gy2d, x2d, 9, n, h, w, stride, 2) \
             if _wgrad2_enabled() \
             else e.conv3x3_wgrad(gy2d, x2d, n, h, w, stride)

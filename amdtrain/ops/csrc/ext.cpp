@@ -82,6 +82,8 @@ at::Tensor conv3x3_wgrad(at::Tensor dy2d, at::Tensor x2d, long Nn, long H,
 at::Tensor tn2_wgrad(at::Tensor dY, at::Tensor X, long taps, long Nn, long H,
                      long W, long stride, long gmode, long kh, long kw,
                      long pad);
+at::Tensor tn2_wgrad_banded(at::Tensor dY, at::Tensor X, long Nn, long H,
+                            long W, long stride);
 at::Tensor tr16_probe(at::Tensor in);
 
 // conv_stem.hip
@@ -154,6 +156,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("taps") = 1, py::arg("Nn") = 0, py::arg("H") = 0,
         py::arg("W") = 0, py::arg("stride") = 1, py::arg("gmode") = 0,
         py::arg("kh") = 3, py::arg("kw") = 3, py::arg("pad") = 1);
+  m.def("tn2_wgrad_banded", &tn2_wgrad_banded);
   m.def("tr16_probe", &tr16_probe);
   m.def("conv_generic_fwd", &conv_generic_fwd);
   m.def("conv_generic_wgrad", &conv_generic_wgrad);

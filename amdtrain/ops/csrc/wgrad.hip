@@ -193,7 +193,10 @@ __device__ __forceinline__ bf16x8 tr_frag(unsigned addr) {
 // the NEWEST chunk's loads at each barrier (counted vmcnt, guide T4) —
 // each wave drains to <= its own newest-chunk loads before the barrier,
 // so the chunk consumed next iteration is complete for every wave.
-template <int GMODE, int NW, int KW, int MC, int NBUF = 2>
+// BAND: group-diagonal wgrad (grouped conv, Cin == N, both % 128): each
+// n-tile needs only ONE k-tile per tap (its own 128-channel window), so
+// tiles = nbn*9 and the output is the COMPACT [N, 9*NCX] band.
+template <int GMODE, int NW, int KW, int MC, int NBUF = 2, bool BAND = false>
 __global__ void __launch_bounds__(TN2_TPB, 2)
 tn2_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
            float* __restrict__ parts, long M, int N, int K9, int Cin,
@@ -213,7 +216,9 @@ tn2_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
   const int tile = bid % tiles;
   const int mpart = bid / tiles;
   const int bn = tile / nbk, bk = tile % nbk;
-  const int n0 = bn * NCY, k0 = bk * NCX;
+  const int n0 = bn * NCY;
+  // BAND: bk is the TAP index; the k window is this n-tile's own channels
+  const int k0 = BAND ? bk * Cin + n0 : bk * NCX;
 
   const long mchunks = (M + MC - 1) / MC;
   const long cpp = (mchunks + msplit - 1) / msplit;
@@ -299,8 +304,10 @@ tn2_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
     }
   }
 
-  // plain (non-atomic) per-mpart partial tile
-  float* out = parts + (long)mpart * N * (long)K9;
+  // plain (non-atomic) per-mpart partial tile; BAND writes the compact
+  // [N, 9*NCX] layout (col = tap*NCX + in-window offset)
+  const int kout_cols = BAND ? 9 * NCX : K9;
+  float* out = parts + (long)mpart * N * (long)kout_cols;
 #pragma unroll
   for (int i = 0; i < 4; ++i)
 #pragma unroll
@@ -308,8 +315,10 @@ tn2_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int n = n0 + wn + i * 16 + fq * 4 + r;
-        const int k = k0 + wk + j * 16 + fr;
-        if (n < N && k < K9) out[(long)n * K9 + k] = acc[i][j][r];
+        const int klocal = wk + j * 16 + fr;
+        const int k = BAND ? bk * NCX + klocal : k0 + klocal;
+        if (n < N && (BAND || k < K9))
+          out[(long)n * kout_cols + k] = acc[i][j][r];
       }
 }
 
@@ -485,6 +494,65 @@ at::Tensor tn2_wgrad(at::Tensor dY, at::Tensor X, long taps, long Nn, long H,
       tn2_launch<1, 2, 2, 64>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
     } else {
       tn2_launch<0, 2, 2, 64>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
+    }
+  }
+  return out;
+}
+
+// Group-diagonal conv3x3 wgrad: returns the COMPACT band [Cout, 9*128]
+// (column = tap*128 + channel offset within the cout-row's own 128-wide
+// group window).  Requires Cin == Cout, both % 128.
+at::Tensor tn2_wgrad_banded(at::Tensor dY, at::Tensor X, long Nn, long H,
+                            long W, long stride) {
+  TORCH_CHECK(dY.is_cuda() && dY.scalar_type() == at::kBFloat16 &&
+              X.scalar_type() == at::kBFloat16);
+  auto Yc = dY.contiguous();
+  auto Xc = X.contiguous();
+  const long M = Yc.size(0);
+  const int N = (int)Yc.size(1);
+  const int Cin = (int)Xc.size(1);
+  TORCH_CHECK(N == Cin && N % 128 == 0);
+  long Hout = (H + 2 - 3) / stride + 1, Wout = (W + 2 - 3) / stride + 1;
+  TORCH_CHECK(M == Nn * Hout * Wout);
+  TnGeom geo{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride, 3, 3, 1};
+  auto out = at::empty({(long)N, (long)9 * 128},
+                       Yc.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const char* tb = std::getenv("AMDTRAIN_TN2_BLOCKS");
+  const int target = tb ? atoi(tb) : 512;
+  const int nbn = N / 128, nbk = 9;
+  const long tiles = (long)nbn * nbk;
+  const long mchunks = (M + 63) / 64;
+  int msplit = (int)std::max<long>(
+      1, std::min<long>(mchunks, target / tiles));
+  auto zp = tn2_zero_page(Yc);
+  at::Tensor parts = out;
+  if (msplit > 1)
+    parts = at::empty({msplit, (long)N * 9 * 128}, out.options());
+  tn2_kernel<2, 2, 2, 64, 2, true>
+      <<<(int)(tiles * msplit), TN2_TPB, 0, stream>>>(
+          (const bf16*)Yc.const_data_ptr(), (const bf16*)Xc.const_data_ptr(),
+          parts.data_ptr<float>(), M, N, 9 * Cin, Cin, geo, nbn, nbk, msplit,
+          (const bf16*)zp.const_data_ptr());
+  CHECK_CUDA_OK();
+  if (msplit > 1) {
+    const long NK = (long)N * 9 * 128;
+    const int nkgrid = amd_grid(NK / 4);
+    if (msplit > 2 * TN2_FOLD) {
+      auto folded = at::empty({TN2_FOLD, NK}, out.options());
+      tn2_collapse_kernel<<<nkgrid * TN2_FOLD, AMD_TPB, 0, stream>>>(
+          (const float*)parts.const_data_ptr(), folded.data_ptr<float>(),
+          msplit, TN2_FOLD, NK);
+      CHECK_CUDA_OK();
+      tn2_collapse_kernel<<<nkgrid, AMD_TPB, 0, stream>>>(
+          (const float*)folded.const_data_ptr(), out.data_ptr<float>(),
+          TN2_FOLD, 1, NK);
+      CHECK_CUDA_OK();
+    } else {
+      tn2_collapse_kernel<<<nkgrid, AMD_TPB, 0, stream>>>(
+          (const float*)parts.const_data_ptr(), out.data_ptr<float>(),
+          msplit, 1, NK);
+      CHECK_CUDA_OK();
     }
   }
   return out;
