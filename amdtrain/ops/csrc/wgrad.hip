@@ -459,8 +459,11 @@ at::Tensor tn2_wgrad(at::Tensor dY, at::Tensor X, long taps, long Nn, long H,
   if (N < 128 && K9 >= 256) {
     // skinny-N conv wgrad: the (2,2,64) tile zero-pads the N half but its
     // MC=64 pipeline beats the fully-dense (1,4,32) config (which is
-    // issue-bound at 16 MFMA/barrier); A/B via AMDTRAIN_TN2_14
-    if (gmode == 2 && std::getenv("AMDTRAIN_TN2_14") == nullptr)
+    // issue-bound at 16 MFMA/barrier) for layer-1-class shapes; the
+    // channel-padded STEM (Cin=8) measured the other way (1.55 vs 1.78 ms
+    // at b512) — its 16x-redundant tap gather prefers the narrower k-tile.
+    // A/B via AMDTRAIN_TN2_14
+    if (gmode == 2 && Cin >= 32 && std::getenv("AMDTRAIN_TN2_14") == nullptr)
       tn2_launch<2, 2, 2, 64>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
     else if (gmode == 2)
       tn2_launch<2, 1, 4, 32>(Yc, Xc, out, M, N, K9, Cin, geo, target, stream);
