@@ -52,6 +52,21 @@ def main(argv=None) -> None:
         run_worker(local_rank, comm.get_world_size(), args, style,
                    global_rank=comm.get_rank())
         return
+    if "OMPI_COMM_WORLD_SIZE" in os.environ \
+            and int(os.environ["OMPI_COMM_WORLD_SIZE"]) > 1:
+        # MPI-style launch (`mpirun -np N python -m ...` — the reference's
+        # `horovodrun -np 4 -H localhost:4` lineage, start.sh:4).  Open MPI
+        # exports the rank/size env vars; rendezvous stays TCP on
+        # --dist-addr/--dist-port (single node: the 127.0.0.1 default works;
+        # multi-node: pass rank 0's address or use `-x MASTER_ADDR`).
+        rank = int(os.environ["OMPI_COMM_WORLD_RANK"])
+        world = int(os.environ["OMPI_COMM_WORLD_SIZE"])
+        local_rank = int(os.environ.get("OMPI_COMM_WORLD_LOCAL_RANK", rank))
+        addr = os.environ.get("MASTER_ADDR", args.dist_addr)
+        comm.init_from_tcp(rank=rank, world_size=world, addr=addr,
+                           port=args.dist_port)
+        run_worker(local_rank, world, args, style, global_rank=rank)
+        return
     nprocs = args.nprocs or torch.cuda.device_count() or 1
     if nprocs == 1:
         run_worker(0, 1, args, style)

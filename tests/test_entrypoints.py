@@ -121,3 +121,32 @@ def test_slurm_rank_math_and_batch_division(monkeypatch):
     assert calls["init"]["world_size"] == 8
     assert calls["global_rank"] == 7
     assert calls["nprocs"] == 4  # run_worker divides batch by this
+
+
+def test_horovod_mpi_style_launch(tmp_path):
+    """MPI-style launch of the horovod entrypoint: two processes with
+    OMPI_COMM_WORLD_* env vars (the `horovodrun -np 2` lineage,
+    reference start.sh:4) rendezvous over TCP and train in lockstep."""
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ,
+                   OMPI_COMM_WORLD_SIZE="2",
+                   OMPI_COMM_WORLD_RANK=str(rank),
+                   OMPI_COMM_WORLD_LOCAL_RANK=str(rank),
+                   PYTHONPATH=root + os.pathsep +
+                   os.environ.get("PYTHONPATH", ""))
+        procs.append(subprocess.Popen(
+            [sys.executable, "-m", "amdtrain.cli.horovod_distributed"]
+            + COMMON + ["--dist-port", "29659"],
+            cwd=str(tmp_path), env=env,
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True))
+    outs = []
+    for p in procs:
+        out, err = p.communicate(timeout=420)
+        outs.append((p.returncode, out, err))
+    for rc, out, err in outs:
+        assert rc == 0, out[-2000:] + err[-2000:]
+    assert (tmp_path / "checkpoint.pth.tar").exists()
