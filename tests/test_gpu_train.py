@@ -313,3 +313,40 @@ def test_ragged_batch_tail_gpu():
     assert torch.isfinite(loss).item()
     assert all(p.grad is not None and torch.isfinite(p.grad).all()
                for p in m.parameters())
+
+
+def test_training_bitwise_deterministic():
+    """Two identical 3-step trainings must produce BITWISE-identical
+    parameters: every kernel on the path (conv fwd/dgrad, tn2 wgrads,
+    BN with fixed-order collapses, fused SGD, CE) has a fixed reduction
+    order — no atomics anywhere in the default path (SURVEY §5
+    determinism; the r1 design needed AMDTRAIN_DETERMINISTIC, r2 does
+    not)."""
+    from amdtrain.models import build_model
+    from amdtrain.ops import CrossEntropyLoss, FusedSGD
+
+    def train_once():
+        torch.manual_seed(42)
+        m = build_model("resnet50", num_classes=100).cuda() \
+            .to(memory_format=torch.channels_last).train()
+        opt = FusedSGD(m.parameters(), lr=0.05, momentum=0.9,
+                       weight_decay=1e-4)
+        crit = CrossEntropyLoss()
+        torch.manual_seed(7)
+        for _ in range(3):
+            x = torch.randn(6, 3, 64, 64, device="cuda") \
+                .contiguous(memory_format=torch.channels_last)
+            t = torch.randint(0, 100, (6,), device="cuda")
+            opt.zero_grad(set_to_none=False)
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                loss = crit(m(x), t)
+            loss.backward()
+            opt.step()
+        torch.cuda.synchronize()
+        return torch.cat([p.detach().reshape(-1).float()
+                          for p in m.parameters()])
+
+    a = train_once()
+    b = train_once()
+    assert torch.equal(a, b), \
+        f"nondeterministic: {(a - b).abs().max().item()}"
