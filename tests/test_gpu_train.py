@@ -350,3 +350,95 @@ def test_training_bitwise_deterministic():
     b = train_once()
     assert torch.equal(a, b), \
         f"nondeterministic: {(a - b).abs().max().item()}"
+
+
+def test_prefetcher_fp16_dtype():
+    """CudaPrefetcher at fp16 output (the O2-fp16 input path; advisor r1
+    found the kernel silently fell back to fp32 here)."""
+    from torch.utils.data import DataLoader, TensorDataset
+    from amdtrain.data.prefetcher import CudaPrefetcher
+
+    imgs = torch.randint(0, 256, (8, 3, 32, 32), dtype=torch.uint8)
+    ds = TensorDataset(imgs, torch.zeros(8, dtype=torch.long))
+    pf = CudaPrefetcher(DataLoader(ds, batch_size=4),
+                        device=torch.device("cuda:0"),
+                        dtype=torch.float16)
+    n = 0
+    for x, t in pf:
+        assert x.dtype == torch.float16 and x.is_cuda
+        assert x.float().abs().max().item() < 4.0  # normalized
+        n += 1
+    assert n == 2
+
+
+def test_eval_mode_inference():
+    """model.eval() + no_grad: the BN eval kernels (running stats) and all
+    conv paths under inference."""
+    from amdtrain.models import build_model
+    torch.manual_seed(0)
+    m = build_model("resnet50").cuda().to(memory_format=torch.channels_last)
+    m.train()
+    x = torch.randn(4, 3, 64, 64, device="cuda") \
+        .contiguous(memory_format=torch.channels_last)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        m(x)  # one train pass to move running stats
+    m.eval()
+    with torch.no_grad(), torch.autocast("cuda", dtype=torch.bfloat16):
+        y1 = m(x)
+        y2 = m(x)
+    assert torch.equal(y1, y2)  # eval is stateless
+    assert torch.isfinite(y1.float()).all()
+
+
+def test_ddp_no_sync_accumulation():
+    """NativeDDP.no_sync gradient accumulation (world 1): two accumulated
+    half-batches equal one full batch."""
+    from amdtrain.models import build_model
+    from amdtrain.parallel import NativeDDP
+
+    torch.manual_seed(0)
+    m = build_model("resnet18", num_classes=10).cuda() \
+        .to(memory_format=torch.channels_last)
+    ddp = NativeDDP(m, bucket_cap_mb=8.0)
+    x = torch.randn(8, 3, 64, 64, device="cuda") \
+        .contiguous(memory_format=torch.channels_last)
+    t = torch.randint(0, 10, (8,), device="cuda")
+
+    def loss_of(xx, tt):
+        from amdtrain.ops import CrossEntropyLoss
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            return CrossEntropyLoss()(ddp(xx), tt)
+
+    ddp.zero_grad()
+    with ddp.no_sync():
+        (loss_of(x[:4], t[:4]) * 0.5).backward()
+    (loss_of(x[4:], t[4:]) * 0.5).backward()
+    acc = [p.grad.detach().float().clone() for p in m.parameters()]
+    ddp.zero_grad()
+    loss_of(x, t).backward()
+    for g_acc, p in zip(acc, m.parameters()):
+        g_full = p.grad.detach().float()
+        lim = 0.08 * g_full.abs().max().item() + 2e-3
+        assert (g_acc - g_full).abs().max().item() <= lim
+
+
+def test_checkpoint_roundtrip_gpu():
+    """Checkpoint save -> load on a GPU bf16/channels_last model restores
+    identical weights (reference dict schema)."""
+    import os
+    from amdtrain.models import build_model
+    from amdtrain.utils import load_checkpoint, save_checkpoint
+    from amdtrain.utils.checkpoint import make_checkpoint_state
+
+    torch.manual_seed(0)
+    m = build_model("resnet18", num_classes=10).cuda() \
+        .to(memory_format=torch.channels_last)
+    state = make_checkpoint_state(0, "resnet18", m, 12.5)
+    path = "/tmp/ck_roundtrip.pth.tar"
+    save_checkpoint(state, False, filename=path)
+    m2 = build_model("resnet18", num_classes=10).cuda()
+    ck = load_checkpoint(path, m2)
+    assert ck["best_acc1"] == 12.5
+    for a, b in zip(m.parameters(), m2.parameters()):
+        assert torch.equal(a.detach().cpu(), b.detach().cpu())
+    os.remove(path)
