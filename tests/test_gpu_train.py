@@ -391,8 +391,11 @@ def test_eval_mode_inference():
 
 
 def test_ddp_no_sync_accumulation():
-    """NativeDDP.no_sync gradient accumulation (world 1): two accumulated
-    half-batches equal one full batch."""
+    """NativeDDP.no_sync gradient accumulation: accumulating two backward
+    passes into the bucket views must equal the SUM of the individual
+    gradients exactly (the stack is bitwise deterministic, so torch.equal
+    applies).  NOTE: half-batch-sum vs full-batch is NOT the invariant
+    here — BatchNorm statistics differ."""
     from amdtrain.models import build_model
     from amdtrain.parallel import NativeDDP
 
@@ -409,17 +412,26 @@ def test_ddp_no_sync_accumulation():
         with torch.autocast("cuda", dtype=torch.bfloat16):
             return CrossEntropyLoss()(ddp(xx), tt)
 
+    # individual grads (BN running stats advance identically in both
+    # schedules, so the per-pass grads are reproducible)
+    sd = {k: v.clone() for k, v in m.state_dict().items()}
+    ddp.zero_grad()
+    loss_of(x[:4], t[:4]).backward()
+    g1 = [p.grad.detach().clone() for p in m.parameters()]
+    ddp.zero_grad()
+    loss_of(x[4:], t[4:]).backward()
+    g2 = [p.grad.detach().clone() for p in m.parameters()]
+
+    # accumulated run from the same starting state
+    m.load_state_dict(sd)
     ddp.zero_grad()
     with ddp.no_sync():
-        (loss_of(x[:4], t[:4]) * 0.5).backward()
-    (loss_of(x[4:], t[4:]) * 0.5).backward()
-    acc = [p.grad.detach().float().clone() for p in m.parameters()]
-    ddp.zero_grad()
-    loss_of(x, t).backward()
-    for g_acc, p in zip(acc, m.parameters()):
-        g_full = p.grad.detach().float()
-        lim = 0.08 * g_full.abs().max().item() + 2e-3
-        assert (g_acc - g_full).abs().max().item() <= lim
+        loss_of(x[:4], t[:4]).backward()
+    loss_of(x[4:], t[4:]).backward()
+    torch.cuda.synchronize()
+    for a, b, p in zip(g1, g2, m.parameters()):
+        assert torch.equal(p.grad.detach(), a + b), \
+            (p.grad.detach() - (a + b)).abs().max().item()
 
 
 def test_checkpoint_roundtrip_gpu():
