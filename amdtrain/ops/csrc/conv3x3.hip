@@ -135,6 +135,27 @@ __device__ __forceinline__ void stage_gathered(
 
 // stage a [128 rows][32 cols] tile of a plain [Rows x ld] matrix at column
 // offset koff (weights)
+// stage ROWS x BKT (ROWS=64 for the NT=64 tile: half the units)
+template <int BKT, int ROWS>
+__device__ __forceinline__ void stage_plain_rows(
+    const bf16* __restrict__ gsrc, long ld, long row0, long rows, long koff,
+    bf16* lds) {
+  const int t = threadIdx.x;
+  constexpr int UPR = BKT / 8;
+  constexpr int UNITS = ROWS * UPR;
+#pragma unroll
+  for (int rnd = 0; rnd < UNITS / GEMM_TPB; ++rnd) {
+    int unit = rnd * GEMM_TPB + t;
+    long row = row0 + unit / UPR;
+    if (row >= rows) row = rows - 1;
+    const bf16* src = gsrc + row * ld + koff + (unit % UPR) * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)(lds + unit * 8), 16,
+        0, 0);
+  }
+}
+
 template <int BKT>
 __device__ __forceinline__ void stage_plain(
     const bf16* __restrict__ gsrc, long ld, long row0, long rows, long koff,
@@ -154,15 +175,18 @@ __device__ __forceinline__ void stage_plain(
   }
 }
 
-// per-block column (sum, sumsq) of the fp32 accumulator tile (see gemm.hip)
+// per-block column (sum, sumsq) of the fp32 accumulator tile (see
+// gemm.hip); NTE = n-tile width (128, or 64 for the banded variant)
+template <int NTE = 128>
 __device__ __forceinline__ void conv_epilogue_stats(
     float* __restrict__ stats, const float* acc_flat, long m0, long M,
     long n0, long N, int bm, int wm, int wn, int fr, int fq,
     bf16* lds_scratch) {
+  constexpr int NF = NTE / 32;  // n fragments per wave
   float* srow = stats + (long)bm * 2 * N;
-  float s1[4], s2[4];
+  float s1[NF], s2[NF];
 #pragma unroll
-  for (int j = 0; j < 4; ++j) {
+  for (int j = 0; j < NF; ++j) {
     s1[j] = 0.f;
     s2[j] = 0.f;
 #pragma unroll
@@ -170,13 +194,13 @@ __device__ __forceinline__ void conv_epilogue_stats(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         long row = m0 + wm + i * 16 + fq * 4 + r;
-        float v = (row < M) ? acc_flat[(i * 4 + j) * 4 + r] : 0.f;
+        float v = (row < M) ? acc_flat[(i * NF + j) * 4 + r] : 0.f;
         s1[j] += v;
         s2[j] += v * v;
       }
   }
 #pragma unroll
-  for (int j = 0; j < 4; ++j) {
+  for (int j = 0; j < NF; ++j) {
 #pragma unroll
     for (int off = 32; off >= 16; off >>= 1) {
       s1[j] += __shfl_down(s1[j], off, AMD_WAVE);
@@ -187,19 +211,19 @@ __device__ __forceinline__ void conv_epilogue_stats(
   __syncthreads();
   if (fq == 0) {
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+    for (int j = 0; j < NF; ++j) {
       int col = wn + j * 16 + fr;
-      red[(wm ? 1 : 0) * 256 + col * 2 + 0] = s1[j];
-      red[(wm ? 1 : 0) * 256 + col * 2 + 1] = s2[j];
+      red[(wm ? 1 : 0) * (2 * NTE) + col * 2 + 0] = s1[j];
+      red[(wm ? 1 : 0) * (2 * NTE) + col * 2 + 1] = s2[j];
     }
   }
   __syncthreads();
   const int t = threadIdx.x;
-  for (int col = t; col < 128; col += GEMM_TPB) {
+  for (int col = t; col < NTE; col += GEMM_TPB) {
     long n = n0 + col;
     if (n < N) {
-      srow[n] = red[col * 2] + red[256 + col * 2];
-      srow[N + n] = red[col * 2 + 1] + red[256 + col * 2 + 1];
+      srow[n] = red[col * 2] + red[2 * NTE + col * 2];
+      srow[N + n] = red[col * 2 + 1] + red[2 * NTE + col * 2 + 1];
     }
   }
 }
@@ -208,10 +232,12 @@ __device__ __forceinline__ void conv_epilogue_stats(
 //   fwd:   A = x rows (AC channels), B = w [NC, 9*AC], C = y [M, NC]
 //   dgrad: A = dy rows (AC = Cout), B = w' [NC = Cin, 9*Cout], C = dx
 // BAND: block-diagonal weight (grouped conv via the dense path, Cin==NC,
-// both % 128): only the K-steps whose 128-channel window matches this
-// n-tile's group window are nonzero — skip the rest (groups/128-per-S of
-// the work)
-template <bool DGRAD, int BKT = BK, bool BAND = false>
+// both % 128): only the K-steps whose NT-channel window matches this
+// n-tile's group window are nonzero — skip the rest.
+// NT: output-tile width (128 default; 64 halves the band window and thus
+// the off-diagonal waste for grouped convs — wave grid becomes 2m x 2n
+// over [128m x 64n], 4x2 fragments per wave).
+template <bool DGRAD, int BKT = BK, bool BAND = false, int NT = 128>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
                bf16* __restrict__ C, long M, int AC, int NC, ConvGeom g,
@@ -222,18 +248,19 @@ conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
 
   const int bid = xcd_swz(blockIdx.x, nbm * nbn);
   const int bm = bid / nbn, bn = bid % nbn;
-  const long m0 = (long)bm * 128, n0 = (long)bn * 128;
+  const long m0 = (long)bm * 128, n0 = (long)bn * NT;
+  constexpr int NFR = NT / 32;  // n fragments per wave (4 at NT=128)
 
   const int t = threadIdx.x;
   const int wave = t / AMD_WAVE, lane = t % AMD_WAVE;
-  const int wm = (wave >> 1) * 64, wn = (wave & 1) * 64;
+  const int wm = (wave >> 1) * 64, wn = (wave & 1) * (NT / 2);
   const int fr = lane & 15, fq = lane >> 4;
 
-  f32x4 acc[4][4];
+  f32x4 acc[4][NFR];
 #pragma unroll
   for (int i = 0; i < 4; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < NFR; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   // hoist the per-unit coordinate decode out of the K-loop
   UnitCoord uc[BKT / 16];
@@ -245,31 +272,35 @@ conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
   }
 
   const int ks_lo = BAND ? (int)(n0 / BKT) : 0;
-  const int ks_hi = BAND ? ks_lo + 128 / BKT : AC / BKT;
+  const int ks_hi = BAND ? ks_lo + NT / BKT : AC / BKT;
   for (int tap = 0; tap < 9; ++tap) {
     const int kh = tap / 3, kw = tap % 3;
     for (int ks = ks_lo; ks < ks_hi; ++ks) {
       const int c0 = ks * BKT;
       __syncthreads();
       stage_gathered<DGRAD, BKT>(A, AC, uc, c0, kh, kw, g, zero_page, As);
-      stage_plain<BKT>(Bw, (long)9 * AC, n0, NC, (long)tap * AC + c0, Bs);
+      if (NT == 128)
+        stage_plain<BKT>(Bw, (long)9 * AC, n0, NC, (long)tap * AC + c0, Bs);
+      else  // 64-row B tile: half the staging rounds
+        stage_plain_rows<BKT, 64>(Bw, (long)9 * AC, n0, NC,
+                                  (long)tap * AC + c0, Bs);
       __syncthreads();
 
 #pragma unroll
       for (int kk = 0; kk < BKT / 32; ++kk) {
-        bf16x8 a[4], b[4];
+        bf16x8 a[4], b[NFR];
 #pragma unroll
         for (int i = 0; i < 4; ++i)
           a[i] = *(const bf16x8*)
               &As[(wm + i * 16 + fr) * BKT + kk * 32 + fq * 8];
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
+        for (int j = 0; j < NFR; ++j)
           b[j] = *(const bf16x8*)
               &Bs[(wn + j * 16 + fr) * BKT + kk * 32 + fq * 8];
 #pragma unroll
         for (int i = 0; i < 4; ++i)
 #pragma unroll
-          for (int j = 0; j < 4; ++j)
+          for (int j = 0; j < NFR; ++j)
             acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 a[i], b[j], acc[i][j], 0, 0, 0);
       }
@@ -279,7 +310,7 @@ conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
 #pragma unroll
   for (int i = 0; i < 4; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j)
+    for (int j = 0; j < NFR; ++j)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         long row = m0 + wm + i * 16 + fq * 4 + r;
@@ -288,8 +319,8 @@ conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
           C[row * NC + col] = __float2bfloat16(acc[i][j][r]);
       }
   if (stats != nullptr)
-    conv_epilogue_stats(stats, (const float*)acc, m0, M, n0, NC, bm, wm, wn,
-                        fr, fq, As);
+    conv_epilogue_stats<NT>(stats, (const float*)acc, m0, M, n0, NC, bm,
+                            wm, wn, fr, fq, As);
 }
 
 // wgrad per tap: dW[cout, tap*Cin + cin] += sum_m dY[m, cout] * Xg[m, cin]
@@ -451,12 +482,16 @@ std::vector<at::Tensor> conv3x3_fwd_stats_impl(at::Tensor x2d, long Nn,
   // A/B (tools/bench_conv3x3.py + in-context traces): BK64 fwd +17% at
   // layer2 (Cin=128, M=400k); layer1 (Cin=64) and layer3/4 (small M)
   // measured WORSE in-context -> Cin>=128 && large-M only
-  if (banded)  // requires Cin == Cout, both % 128 (host-checked)
-    conv3x3_kernel<false, BK, true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
-        (const bf16*)x2d.const_data_ptr(), (const bf16*)w2d.const_data_ptr(),
-        (bf16*)y.data_ptr(), M, (int)Cin, (int)Cout, g, nbm, nbn,
-        (const bf16*)zp.const_data_ptr(), stats_ptr);
-  else if (bk64 && Cin % 64 == 0 && Cin >= 128 && M >= 200000)
+  if (banded) {  // requires Cin == Cout, both % 128 (host-checked)
+    // NT=64 tile halves the group-band window (and the off-diag waste)
+    int nbn64 = (int)((Cout + 63) / 64);
+    conv3x3_kernel<false, BK, true, 64>
+        <<<nbm * nbn64, GEMM_TPB, 0, stream>>>(
+            (const bf16*)x2d.const_data_ptr(),
+            (const bf16*)w2d.const_data_ptr(), (bf16*)y.data_ptr(), M,
+            (int)Cin, (int)Cout, g, nbm, nbn64,
+            (const bf16*)zp.const_data_ptr(), stats_ptr);
+  } else if (bk64 && Cin % 64 == 0 && Cin >= 128 && M >= 200000)
     conv3x3_kernel<false, 64><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)x2d.const_data_ptr(), (const bf16*)w2d.const_data_ptr(),
         (bf16*)y.data_ptr(), M, (int)Cin, (int)Cout, g, nbm, nbn,
@@ -506,11 +541,15 @@ at::Tensor conv3x3_dgrad(at::Tensor dy2d, long Nn, long H, long W,
   auto zp = zero_page_for(dy2d);
   if (banded) {
     TORCH_CHECK(Cin == Cout && Cin % 128 == 0);
-    conv3x3_kernel<true, BK, true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
-        (const bf16*)dy2d.const_data_ptr(),
-        (const bf16*)wrot.const_data_ptr(), (bf16*)dx.data_ptr(), M,
-        (int)Cout, (int)Cin, g, nbm, nbn,
-        (const bf16*)zp.const_data_ptr(), nullptr);
+    // NT=64 tile: the group-diagonal K window shrinks to 64 channels,
+    // halving the off-diagonal waste vs the 128-wide tile
+    int nbn64 = (int)(Cin / 64);
+    conv3x3_kernel<true, BK, true, 64>
+        <<<nbm * nbn64, GEMM_TPB, 0, stream>>>(
+            (const bf16*)dy2d.const_data_ptr(),
+            (const bf16*)wrot.const_data_ptr(), (bf16*)dx.data_ptr(), M,
+            (int)Cout, (int)Cin, g, nbm, nbn64,
+            (const bf16*)zp.const_data_ptr(), nullptr);
     CHECK_CUDA_OK();
     return dx;
   }
