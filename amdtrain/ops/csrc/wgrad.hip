@@ -110,7 +110,7 @@ constexpr int tile_bytes() {
 //   m' = 8*((lane>>3)&3) + 4*(lane>>5) + ((lane>>1)&3),
 //   c  = (lane&1)*8 .. +8
 // (inverse of st_byte: lane*16 = r*512 + q*128 + j*32 + c_half*2).
-template <int MODE, int NCOLS, int MC>
+template <int MODE, int NCOLS, int MC, int NWAVES = 4>
 __device__ __forceinline__ void stage_tn2(
     const bf16* __restrict__ g, int ld, long m0, long M, int col0,
     int total_cols, int cin, const TnGeom& geo, const bf16* __restrict__ zp,
@@ -122,8 +122,8 @@ __device__ __forceinline__ void stage_tn2(
   const int c_half = (lane & 1) * 8;
   const int mfrag = q * 8 + r * 4 + j;  // m offset within the 32-row block
 #pragma unroll
-  for (int ss = 0; ss < SUBTILES / 4; ++ss) {
-    const int s = ss * 4 + wave;
+  for (int ss = 0; ss < SUBTILES / NWAVES; ++ss) {
+    const int s = ss * NWAVES + wave;
     const int mb = s / (NCOLS / 16);         // 32-row block
     const int cs = (s % (NCOLS / 16)) * 16;  // subtile column base
     const int mloc = mb * 32 + mfrag;
@@ -159,6 +159,8 @@ __device__ __forceinline__ void vmcnt_wait() {
   else if constexpr (N == 8) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
   else if constexpr (N == 10)
     asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+  else if constexpr (N == 12)
+    asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
   else if constexpr (N == 16)
     asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
   else static_assert(N == 0, "unsupported vmcnt");
@@ -197,17 +199,18 @@ __device__ __forceinline__ bf16x8 tr_frag(unsigned addr) {
 // n-tile needs only ONE k-tile per tap (its own 128-channel window), so
 // tiles = nbn*9 and the output is the COMPACT [N, 9*NCX] band.
 template <int GMODE, int NW, int KW, int MC, int NBUF = 2, bool BAND = false>
-__global__ void __launch_bounds__(TN2_TPB, 2)
+__global__ void __launch_bounds__(NW * KW * 64, (NW * KW == 4 ? 2 : 1))
 tn2_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
            float* __restrict__ parts, long M, int N, int K9, int Cin,
            TnGeom geo, int nbn, int nbk, int msplit,
            const bf16* __restrict__ zp) {
+  constexpr int NWAVES = NW * KW;  // 4 (256 thr, 2 blk/CU) or 8 (512, 1)
   constexpr int NCY = NW * 64;   // dY tile cols
   constexpr int NCX = KW * 64;   // X tile cols
   constexpr int KSTEPS = MC / 32;
   // per-thread global_load_lds issues per chunk (both operands):
-  // subtiles = (MC/32)*(cols/16) per operand, one wave-round each -> /4
-  constexpr int LC = (MC / 32) * (NCY / 16 + NCX / 16) / 4;
+  // subtiles = (MC/32)*(cols/16) per operand, one wave-round each
+  constexpr int LC = (MC / 32) * (NCY / 16 + NCX / 16) * 64 / (NW * KW * 64);
   __shared__ __align__(16) char Ys[NBUF][tile_bytes<NCY, MC>()];
   __shared__ __align__(16) char Xs[NBUF][tile_bytes<NCX, MC>()];
 
@@ -248,14 +251,15 @@ tn2_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
     constexpr unsigned YB = tile_bytes<NCY, MC>();
     constexpr unsigned XB = tile_bytes<NCX, MC>();
 
-    stage_tn2<0, NCY, MC>(dY, N, mc0 * MC, M, n0, N, Cin, geo, zp, (bf16*)Ys[0]);
-    stage_tn2<GMODE, NCX, MC>(X, Cin, mc0 * MC, M, k0, K9, Cin, geo, zp,
-                              (bf16*)Xs[0]);
+    stage_tn2<0, NCY, MC, NWAVES>(dY, N, mc0 * MC, M, n0, N, Cin, geo, zp,
+                                  (bf16*)Ys[0]);
+    stage_tn2<GMODE, NCX, MC, NWAVES>(X, Cin, mc0 * MC, M, k0, K9, Cin, geo,
+                                      zp, (bf16*)Xs[0]);
     if (NBUF >= 3 && mc0 + 1 < mc1) {
-      stage_tn2<0, NCY, MC>(dY, N, (mc0 + 1) * MC, M, n0, N, Cin, geo, zp,
-                            (bf16*)Ys[1]);
-      stage_tn2<GMODE, NCX, MC>(X, Cin, (mc0 + 1) * MC, M, k0, K9, Cin, geo,
-                                zp, (bf16*)Xs[1]);
+      stage_tn2<0, NCY, MC, NWAVES>(dY, N, (mc0 + 1) * MC, M, n0, N, Cin,
+                                    geo, zp, (bf16*)Ys[1]);
+      stage_tn2<GMODE, NCX, MC, NWAVES>(X, Cin, (mc0 + 1) * MC, M, k0, K9,
+                                        Cin, geo, zp, (bf16*)Xs[1]);
       vmcnt_wait<LC>();  // chunk 0 complete; chunk 1 may stay in flight
     } else {
       vmcnt_wait<0>();
@@ -267,10 +271,10 @@ tn2_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
       const long pre = NBUF >= 3 ? mc + 2 : mc + 1;
       if (pre < mc1) {
         const int nxt = (int)((pre - mc0) % NBUF);
-        stage_tn2<0, NCY, MC>(dY, N, pre * MC, M, n0, N, Cin, geo, zp,
-                              (bf16*)Ys[nxt]);
-        stage_tn2<GMODE, NCX, MC>(X, Cin, pre * MC, M, k0, K9, Cin, geo,
-                                  zp, (bf16*)Xs[nxt]);
+        stage_tn2<0, NCY, MC, NWAVES>(dY, N, pre * MC, M, n0, N, Cin, geo,
+                                      zp, (bf16*)Ys[nxt]);
+        stage_tn2<GMODE, NCX, MC, NWAVES>(X, Cin, pre * MC, M, k0, K9, Cin,
+                                          geo, zp, (bf16*)Xs[nxt]);
       }
       const unsigned yb = ybase0 + cur * YB;
       const unsigned xb = xbase0 + cur * XB;
@@ -389,7 +393,7 @@ void tn2_launch(const at::Tensor& dY, const at::Tensor& X, at::Tensor& out,
   if (msplit > 1)
     parts = at::empty({msplit, (long)N * K9}, out.options());
   tn2_kernel<GMODE, NW, KW, MC, NBUF>
-      <<<(int)(tiles * msplit), TN2_TPB, 0, stream>>>(
+      <<<(int)(tiles * msplit), NW * KW * 64, 0, stream>>>(
           (const bf16*)dY.const_data_ptr(), (const bf16*)X.const_data_ptr(),
           parts.data_ptr<float>(), M, N, K9, Cin, geo, nbn, nbk, msplit,
           (const bf16*)zp.const_data_ptr());
@@ -478,7 +482,22 @@ at::Tensor tn2_wgrad(at::Tensor dY, at::Tensor X, long taps, long Nn, long H,
       const char* v = std::getenv("AMDTRAIN_TN2_PIPE3");
       return v && v[0] == '1';
     }();
-    if (pipe3) {
+    static const bool wide8 = []() {  // A/B override
+      const char* v = std::getenv("AMDTRAIN_TN2_W8");
+      return v && v[0] == '1';
+    }();
+    // 8-wave (2m x 4k) block halves dY re-reads (two k-tiles per staged
+    // chunk); measured +11% at deep-K (K9=4608 layer4 3x3), flat at
+    // smaller K (L3 already absorbs those re-reads) -> auto for K>=4096
+    if ((wide8 || (gmode == 2 && K9 >= 4096)) && K9 % 256 == 0
+        && gmode != 1) {
+      if (gmode == 2)
+        tn2_launch<2, 2, 4, 64>(Yc, Xc, out, M, N, K9, Cin, geo, target,
+                                stream);
+      else
+        tn2_launch<0, 2, 4, 64>(Yc, Xc, out, M, N, K9, Cin, geo, target,
+                                stream);
+    } else if (pipe3) {
       // A/B: 3-buffer counted-vmcnt pipeline at MC=32 (48 KB LDS -> 3
       // blocks/CU).  MEASURED -10..-15% vs the MC=64 double buffer
       // (docs/KERNELS.md negative-results) — kept off-by-default
