@@ -95,17 +95,18 @@ def test_engine_train_loop_gpu():
     assert isinstance(acc, float)
 
 
-def test_residual_grad_fusion_parity():
-    """Identity-shortcut gradient fused into conv1's dgrad epilogue
+@pytest.mark.parametrize("arch", ["resnet50", "resnet18"])
+def test_residual_grad_fusion_parity(arch):
+    """Shortcut gradient rerouted into the producing BN's backward
     (ResidualGradTap) must give the same grads as plain autograd
-    accumulation (AMDTRAIN_RESFUSE=0)."""
+    accumulation (AMDTRAIN_RESFUSE=0) — both Bottleneck and BasicBlock."""
     import os
     from amdtrain.models import build_model
 
     def run(fuse):
         os.environ["AMDTRAIN_RESFUSE"] = "1" if fuse else "0"
         torch.manual_seed(0)
-        m = build_model("resnet50").cuda() \
+        m = build_model(arch).cuda() \
             .to(memory_format=torch.channels_last).train()
         x = torch.randn(4, 3, 64, 64, device="cuda") \
             .contiguous(memory_format=torch.channels_last)
