@@ -1,3 +1,6 @@
+"""A/B the stem (7x7 s2 Cin=8-padded) wgrad tn2 route on hardware:
+default (1,4,32) vs AMDTRAIN_TN2_14 variants; used to pin the per-shape
+config choice (see wgrad.hip dispatch comments)."""
 import sys, os, time
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
