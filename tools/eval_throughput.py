@@ -9,9 +9,13 @@ on-GPU uint8 normalize -> bf16 autocast NHWC forward.
 
 import argparse
 import json
+import os
+import sys
 import time
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
 def main():
