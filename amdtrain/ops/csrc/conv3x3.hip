@@ -237,14 +237,20 @@ __device__ __forceinline__ void conv_epilogue_stats(
 // NT: output-tile width (128 default; 64 halves the band window and thus
 // the off-diagonal waste for grouped convs — wave grid becomes 2m x 2n
 // over [128m x 64n], 4x2 fragments per wave).
-template <bool DGRAD, int BKT = BK, bool BAND = false, int NT = 128>
+// DB: double-buffered (tap, ks) pipeline — stage iteration it+1's A/B
+// tiles while the MFMAs consume iteration it's (one barrier per iteration
+// instead of two; the barrier's implicit vmcnt(0) drains the DMA queue),
+// mirroring the gemm.hip DB loop.
+template <bool DGRAD, int BKT = BK, bool BAND = false, int NT = 128,
+          bool DB = false>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
                bf16* __restrict__ C, long M, int AC, int NC, ConvGeom g,
                int nbm, int nbn, const bf16* __restrict__ zero_page,
                float* __restrict__ stats) {
-  __shared__ bf16 As[128 * BKT];
-  __shared__ bf16 Bs[128 * BKT];
+  __shared__ bf16 SMEM[(DB ? 2 : 1) * 2 * 128 * BKT];
+  bf16* const As = SMEM;
+  bf16* const Bs = SMEM + 128 * BKT;
 
   const int bid = xcd_swz(blockIdx.x, nbm * nbn);
   const int bm = bid / nbn, bn = bid % nbn;
@@ -273,37 +279,56 @@ conv3x3_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bw,
 
   const int ks_lo = BAND ? (int)(n0 / BKT) : 0;
   const int ks_hi = BAND ? ks_lo + NT / BKT : AC / BKT;
-  for (int tap = 0; tap < 9; ++tap) {
-    const int kh = tap / 3, kw = tap % 3;
-    for (int ks = ks_lo; ks < ks_hi; ++ks) {
-      const int c0 = ks * BKT;
-      __syncthreads();
-      stage_gathered<DGRAD, BKT>(A, AC, uc, c0, kh, kw, g, zero_page, As);
-      if (NT == 128)
-        stage_plain<BKT>(Bw, (long)9 * AC, n0, NC, (long)tap * AC + c0, Bs);
-      else  // 64-row B tile: half the staging rounds
-        stage_plain_rows<BKT, 64>(Bw, (long)9 * AC, n0, NC,
-                                  (long)tap * AC + c0, Bs);
-      __syncthreads();
-
+  const int nks = ks_hi - ks_lo;
+  const int total = 9 * nks;
+  auto stage_it = [&](int it, bf16* as) {
+    const int tap = it / nks, ks = ks_lo + it % nks;
+    const int kh = tap / 3, kw = tap % 3, c0 = ks * BKT;
+    stage_gathered<DGRAD, BKT>(A, AC, uc, c0, kh, kw, g, zero_page, as);
+    if (NT == 128)
+      stage_plain<BKT>(Bw, (long)9 * AC, n0, NC, (long)tap * AC + c0,
+                       as + 128 * BKT);
+    else  // 64-row B tile: half the staging rounds
+      stage_plain_rows<BKT, 64>(Bw, (long)9 * AC, n0, NC,
+                                (long)tap * AC + c0, as + 128 * BKT);
+  };
+  auto compute_it = [&](const bf16* as) {
+    const bf16* bs = as + 128 * BKT;
 #pragma unroll
-      for (int kk = 0; kk < BKT / 32; ++kk) {
-        bf16x8 a[4], b[NFR];
+    for (int kk = 0; kk < BKT / 32; ++kk) {
+      bf16x8 a[4], b[NFR];
 #pragma unroll
-        for (int i = 0; i < 4; ++i)
-          a[i] = *(const bf16x8*)
-              &As[(wm + i * 16 + fr) * BKT + kk * 32 + fq * 8];
+      for (int i = 0; i < 4; ++i)
+        a[i] = *(const bf16x8*)
+            &as[(wm + i * 16 + fr) * BKT + kk * 32 + fq * 8];
+#pragma unroll
+      for (int j = 0; j < NFR; ++j)
+        b[j] = *(const bf16x8*)
+            &bs[(wn + j * 16 + fr) * BKT + kk * 32 + fq * 8];
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
 #pragma unroll
         for (int j = 0; j < NFR; ++j)
-          b[j] = *(const bf16x8*)
-              &Bs[(wn + j * 16 + fr) * BKT + kk * 32 + fq * 8];
-#pragma unroll
-        for (int i = 0; i < 4; ++i)
-#pragma unroll
-          for (int j = 0; j < NFR; ++j)
-            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a[i], b[j], acc[i][j], 0, 0, 0);
-      }
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+  };
+  if (DB) {
+    constexpr int HB = 2 * 128 * BKT;
+    if (total > 0) stage_it(0, SMEM);
+    for (int it = 0; it < total; ++it) {
+      bf16* const as = SMEM + (it & 1) * HB;
+      __syncthreads();  // implicit vmcnt(0) drains this iter's DMA; all
+                        // waves are past reading the other buffer
+      if (it + 1 < total) stage_it(it + 1, SMEM + ((it + 1) & 1) * HB);
+      compute_it(as);
+    }
+  } else {
+    for (int it = 0; it < total; ++it) {
+      __syncthreads();
+      stage_it(it, SMEM);
+      __syncthreads();
+      compute_it(SMEM);
     }
   }
 
@@ -444,6 +469,16 @@ rotate_weight_kernel(const bf16* __restrict__ w, bf16* __restrict__ out,
 
 }  // namespace
 
+// AMDTRAIN_CONV3X3_DB=0 forces the single-buffer (tap, ks) loop; default
+// is the double-buffered pipeline (see kernel comment / gemm.hip).
+static bool conv3x3_db_enabled() {
+  static const bool v = []() {
+    const char* e = std::getenv("AMDTRAIN_CONV3X3_DB");
+    return !(e && e[0] == '0');
+  }();
+  return v;
+}
+
 // host-side 16B zero page (device memory), one per device, created lazily
 static at::Tensor zero_page_for(const at::Tensor& like) {
   static thread_local at::Tensor zp;
@@ -485,17 +520,41 @@ std::vector<at::Tensor> conv3x3_fwd_stats_impl(at::Tensor x2d, long Nn,
   if (banded) {  // requires Cin == Cout, both % 128 (host-checked)
     // NT=64 tile halves the group-band window (and the off-diag waste)
     int nbn64 = (int)((Cout + 63) / 64);
-    conv3x3_kernel<false, BK, true, 64>
-        <<<nbm * nbn64, GEMM_TPB, 0, stream>>>(
+    if (conv3x3_db_enabled())
+      conv3x3_kernel<false, BK, true, 64, true>
+          <<<nbm * nbn64, GEMM_TPB, 0, stream>>>(
+              (const bf16*)x2d.const_data_ptr(),
+              (const bf16*)w2d.const_data_ptr(), (bf16*)y.data_ptr(), M,
+              (int)Cin, (int)Cout, g, nbm, nbn64,
+              (const bf16*)zp.const_data_ptr(), stats_ptr);
+    else
+      conv3x3_kernel<false, BK, true, 64>
+          <<<nbm * nbn64, GEMM_TPB, 0, stream>>>(
+              (const bf16*)x2d.const_data_ptr(),
+              (const bf16*)w2d.const_data_ptr(), (bf16*)y.data_ptr(), M,
+              (int)Cin, (int)Cout, g, nbm, nbn64,
+              (const bf16*)zp.const_data_ptr(), stats_ptr);
+  } else if (bk64 && Cin % 64 == 0 && Cin >= 128 && M >= 200000) {
+    if (conv3x3_db_enabled())
+      conv3x3_kernel<false, 64, false, 128, true>
+          <<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+              (const bf16*)x2d.const_data_ptr(),
+              (const bf16*)w2d.const_data_ptr(), (bf16*)y.data_ptr(), M,
+              (int)Cin, (int)Cout, g, nbm, nbn,
+              (const bf16*)zp.const_data_ptr(), stats_ptr);
+    else
+      conv3x3_kernel<false, 64><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+          (const bf16*)x2d.const_data_ptr(),
+          (const bf16*)w2d.const_data_ptr(), (bf16*)y.data_ptr(), M,
+          (int)Cin, (int)Cout, g, nbm, nbn,
+          (const bf16*)zp.const_data_ptr(), stats_ptr);
+  } else if (conv3x3_db_enabled())
+    conv3x3_kernel<false, BK, false, 128, true>
+        <<<nbm * nbn, GEMM_TPB, 0, stream>>>(
             (const bf16*)x2d.const_data_ptr(),
             (const bf16*)w2d.const_data_ptr(), (bf16*)y.data_ptr(), M,
-            (int)Cin, (int)Cout, g, nbm, nbn64,
+            (int)Cin, (int)Cout, g, nbm, nbn,
             (const bf16*)zp.const_data_ptr(), stats_ptr);
-  } else if (bk64 && Cin % 64 == 0 && Cin >= 128 && M >= 200000)
-    conv3x3_kernel<false, 64><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
-        (const bf16*)x2d.const_data_ptr(), (const bf16*)w2d.const_data_ptr(),
-        (bf16*)y.data_ptr(), M, (int)Cin, (int)Cout, g, nbm, nbn,
-        (const bf16*)zp.const_data_ptr(), stats_ptr);
   else
     conv3x3_kernel<false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)x2d.const_data_ptr(), (const bf16*)w2d.const_data_ptr(),
@@ -544,12 +603,20 @@ at::Tensor conv3x3_dgrad(at::Tensor dy2d, long Nn, long H, long W,
     // NT=64 tile: the group-diagonal K window shrinks to 64 channels,
     // halving the off-diagonal waste vs the 128-wide tile
     int nbn64 = (int)(Cin / 64);
-    conv3x3_kernel<true, BK, true, 64>
-        <<<nbm * nbn64, GEMM_TPB, 0, stream>>>(
-            (const bf16*)dy2d.const_data_ptr(),
-            (const bf16*)wrot.const_data_ptr(), (bf16*)dx.data_ptr(), M,
-            (int)Cout, (int)Cin, g, nbm, nbn64,
-            (const bf16*)zp.const_data_ptr(), nullptr);
+    if (conv3x3_db_enabled())
+      conv3x3_kernel<true, BK, true, 64, true>
+          <<<nbm * nbn64, GEMM_TPB, 0, stream>>>(
+              (const bf16*)dy2d.const_data_ptr(),
+              (const bf16*)wrot.const_data_ptr(), (bf16*)dx.data_ptr(), M,
+              (int)Cout, (int)Cin, g, nbm, nbn64,
+              (const bf16*)zp.const_data_ptr(), nullptr);
+    else
+      conv3x3_kernel<true, BK, true, 64>
+          <<<nbm * nbn64, GEMM_TPB, 0, stream>>>(
+              (const bf16*)dy2d.const_data_ptr(),
+              (const bf16*)wrot.const_data_ptr(), (bf16*)dx.data_ptr(), M,
+              (int)Cout, (int)Cin, g, nbm, nbn64,
+              (const bf16*)zp.const_data_ptr(), nullptr);
     CHECK_CUDA_OK();
     return dx;
   }
@@ -563,6 +630,13 @@ at::Tensor conv3x3_dgrad(at::Tensor dy2d, long Nn, long H, long W,
         (const bf16*)wrot.const_data_ptr(), (bf16*)dx.data_ptr(), M,
         (int)Cout, (int)Cin, g, nbm, nbn,
         (const bf16*)zp.const_data_ptr(), nullptr);
+  else if (conv3x3_db_enabled())
+    conv3x3_kernel<true, BK, false, 128, true>
+        <<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+            (const bf16*)dy2d.const_data_ptr(),
+            (const bf16*)wrot.const_data_ptr(), (bf16*)dx.data_ptr(), M,
+            (int)Cout, (int)Cin, g, nbm, nbn,
+            (const bf16*)zp.const_data_ptr(), nullptr);
   else
     conv3x3_kernel<true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)dy2d.const_data_ptr(),

@@ -82,6 +82,23 @@ __device__ __forceinline__ void stage_tile_rows(
   }
 }
 
+// Double-buffered K-loop variants: the single-buffer loop serializes DMA
+// and MFMA within a block and leans on the 2-blocks/CU interleave for
+// overlap.  Measured (b512 shape sweep + e2e): +1-9% on the streaming
+// 1x1 shapes, -15% on the deep-K small-M shapes (K=2048: 64 k-steps of
+// compute-heavy reuse where the doubled LDS hurts more than overlap
+// helps) -> ON by default for K <= GEMM_DB_KMAX, AMDTRAIN_GEMM_DB=0/1
+// forces either path.
+constexpr long GEMM_DB_KMAX = 1024;
+inline bool gemm_db_enabled(long K) {
+  static const int v = []() {
+    const char* e = std::getenv("AMDTRAIN_GEMM_DB");
+    return e ? (e[0] == '1' ? 1 : 0) : -1;
+  }();
+  if (v >= 0) return v == 1;
+  return K <= GEMM_DB_KMAX;
+}
+
 // compact output row m=(n,ho,wo) -> strided input row (n, ho*s, wo*s)
 struct StrideMap {
   int H, W, Hout, Wout, stride;  // input spatial dims + output dims
@@ -163,13 +180,19 @@ __device__ __forceinline__ void epilogue_stats(
 // of the 2 B scattered stores); kept env-gated as a documented negative.
 // LDSE=1: stage the C tile through LDS and emit coalesced 16 B stores
 // instead of per-element column-strided 2 B stores.
-template <bool F32OUT, bool STRIDED, bool NT = false, bool LDSE = false>
+// DB=1: double-buffered K-loop — stage chunk kt+1 while the MFMAs consume
+// chunk kt (one barrier per chunk instead of two, per-block load/compute
+// overlap instead of relying on the 2-blocks/CU interleave drifting into
+// opposite phase).  The barrier's implicit vmcnt(0) wait is what drains
+// the global_load_lds queue, exactly as in the single-buffer path.
+template <bool F32OUT, bool STRIDED, bool NT = false, bool LDSE = false,
+          bool DB = false>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                void* __restrict__ C, long M, long N, long K, int nbm,
                int nbn, StrideMap sm, float* __restrict__ stats) {
-  __shared__ bf16 SMEM[(BM + BN) * BK];  // As | Bs (contiguous: the LDSE
-  bf16* const As = SMEM;                 // epilogue reuses all 16 KB)
+  __shared__ bf16 SMEM[(DB ? 2 : 1) * (BM + BN) * BK];  // As | Bs
+  bf16* const As = SMEM;    // (contiguous: the LDSE epilogue reuses 16 KB)
   bf16* const Bs = SMEM + BM * BK;
 
   const int nwg = nbm * nbn;
@@ -202,28 +225,47 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
   }
 
   const long ksteps = K / BK;
-  for (long kt = 0; kt < ksteps; ++kt) {
-    __syncthreads();  // previous compute done before overwriting LDS
+  auto stage = [&](long kt, bf16* as) {
     if (STRIDED)
-      stage_tile_rows(A, K, arow, kt * BK, As);
+      stage_tile_rows(A, K, arow, kt * BK, as);
     else
-      stage_tile_128x32(A, K, m0, M, kt * BK, As);
-    stage_tile_128x32(B, K, n0, N, kt * BK, Bs);
-    __syncthreads();  // barrier drains the global_load_lds queue
-
+      stage_tile_128x32(A, K, m0, M, kt * BK, as);
+    stage_tile_128x32(B, K, n0, N, kt * BK, as + BM * BK);
+  };
+  auto compute = [&](const bf16* as) {
+    const bf16* bs = as + BM * BK;
     bf16x8 a[4], b[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
-      a[i] = *(const bf16x8*)&As[(wm + i * 16 + fr) * BK + fq * 8];
+      a[i] = *(const bf16x8*)&as[(wm + i * 16 + fr) * BK + fq * 8];
 #pragma unroll
     for (int j = 0; j < 4; ++j)
-      b[j] = *(const bf16x8*)&Bs[(wn + j * 16 + fr) * BK + fq * 8];
+      b[j] = *(const bf16x8*)&bs[(wn + j * 16 + fr) * BK + fq * 8];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
 #pragma unroll
       for (int j = 0; j < 4; ++j)
         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a[i], b[j], acc[i][j], 0, 0, 0);
+  };
+  if (DB) {
+    constexpr int HB = (BM + BN) * BK;
+    if (ksteps > 0) stage(0, SMEM);
+    for (long kt = 0; kt < ksteps; ++kt) {
+      bf16* const as = SMEM + (kt & 1) * HB;
+      __syncthreads();  // drains this chunk's DMA queue (implicit
+                        // vmcnt(0)); all waves are past reading the
+                        // other buffer, which kt+1 overwrites below
+      if (kt + 1 < ksteps) stage(kt + 1, SMEM + ((kt + 1) & 1) * HB);
+      compute(as);
+    }
+  } else {
+    for (long kt = 0; kt < ksteps; ++kt) {
+      __syncthreads();  // previous compute done before overwriting LDS
+      stage(kt, SMEM);
+      __syncthreads();  // barrier drains the global_load_lds queue
+      compute(SMEM);
+    }
   }
 
   // epilogue: C[m0+wm+i*16+fq*4+r][n0+wn+j*16+fr]
@@ -291,12 +333,14 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
 // uniform streaming-efficiency wall, not the A re-reads — the rr-floor
 // coincidence in the first analysis was misleading); kept because it also
 // halves launch width and is covered by tests.
+template <bool DB = false>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 gemm_bt_n2_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                   bf16* __restrict__ C, long M, long N, long K, int nbm,
                   int nbn2, float* __restrict__ stats, int nbm_stats) {
-  __shared__ bf16 As[BM * BK];
-  __shared__ bf16 Bs[256 * BK];
+  __shared__ bf16 SMEM[(DB ? 2 : 1) * (BM + 256) * BK];
+  bf16* const As = SMEM;
+  bf16* const Bs = SMEM + BM * BK;
 
   const int bid = xcd_swizzle(blockIdx.x, nbm * nbn2);
   const int bm = bid / nbn2, bn = bid % nbn2;
@@ -317,21 +361,22 @@ gemm_bt_n2_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
     for (int j = 0; j < 8; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const long ksteps = K / BK;
-  for (long kt = 0; kt < ksteps; ++kt) {
-    __syncthreads();
-    stage_tile_128x32(A, K, m0, M, kt * BK, As);
-    stage_tile_128x32(B, K, n0, N, kt * BK, Bs);
-    stage_tile_128x32(B, K, n0 + 128, N, kt * BK, Bs + 128 * BK);
-    __syncthreads();
-
+  auto stage = [&](long kt, bf16* as) {
+    bf16* bs = as + BM * BK;
+    stage_tile_128x32(A, K, m0, M, kt * BK, as);
+    stage_tile_128x32(B, K, n0, N, kt * BK, bs);
+    stage_tile_128x32(B, K, n0 + 128, N, kt * BK, bs + 128 * BK);
+  };
+  auto compute = [&](const bf16* as) {
+    const bf16* bs = as + BM * BK;
     bf16x8 a[4], b[8];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
-      a[i] = *(const bf16x8*)&As[(wm + i * 16 + fr) * BK + fq * 8];
+      a[i] = *(const bf16x8*)&as[(wm + i * 16 + fr) * BK + fq * 8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const int jn = (j < 4) ? wn + j * 16 : 128 + wn + (j - 4) * 16;
-      b[j] = *(const bf16x8*)&Bs[(jn + fr) * BK + fq * 8];
+      b[j] = *(const bf16x8*)&bs[(jn + fr) * BK + fq * 8];
     }
 #pragma unroll
     for (int i = 0; i < 4; ++i)
@@ -339,6 +384,23 @@ gemm_bt_n2_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
       for (int j = 0; j < 8; ++j)
         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a[i], b[j], acc[i][j], 0, 0, 0);
+  };
+  if (DB) {
+    constexpr int HB = (BM + 256) * BK;
+    if (ksteps > 0) stage(0, SMEM);
+    for (long kt = 0; kt < ksteps; ++kt) {
+      bf16* const as = SMEM + (kt & 1) * HB;
+      __syncthreads();  // implicit vmcnt(0) drains this chunk's DMA
+      if (kt + 1 < ksteps) stage(kt + 1, SMEM + ((kt + 1) & 1) * HB);
+      compute(as);
+    }
+  } else {
+    for (long kt = 0; kt < ksteps; ++kt) {
+      __syncthreads();
+      stage(kt, SMEM);
+      __syncthreads();
+      compute(SMEM);
+    }
   }
 
 #pragma unroll
@@ -662,9 +724,16 @@ at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out,
     // wide-N small-K: dual-n-tile variant (see kernel comment; measured
     // perf-neutral, halves nominal A traffic and launch width)
     int nbn2 = nbn / 2;
-    gemm_bt_n2_kernel<<<nbm * nbn2, GEMM_TPB, 0, stream>>>(
-        (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-        (bf16*)C.data_ptr(), M, N, K, nbm, nbn2, nullptr, nbm);
+    if (gemm_db_enabled(K))
+      gemm_bt_n2_kernel<true><<<nbm * nbn2, GEMM_TPB, 0, stream>>>(
+          (const bf16*)Ac.const_data_ptr(),
+          (const bf16*)Bc.const_data_ptr(), (bf16*)C.data_ptr(), M, N, K,
+          nbm, nbn2, nullptr, nbm);
+    else
+      gemm_bt_n2_kernel<<<nbm * nbn2, GEMM_TPB, 0, stream>>>(
+          (const bf16*)Ac.const_data_ptr(),
+          (const bf16*)Bc.const_data_ptr(), (bf16*)C.data_ptr(), M, N, K,
+          nbm, nbn2, nullptr, nbm);
     CHECK_CUDA_OK();
     return C;
   }
@@ -687,6 +756,7 @@ at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out,
     const char* v = std::getenv("AMDTRAIN_GEMM_NT");
     return v && v[0] == '1';
   }();
+  const bool use_db = gemm_db_enabled(K);
   static const bool use_ldse = []() {
     const char* v = std::getenv("AMDTRAIN_GEMM_LDSE");
     return v && v[0] == '1';
@@ -705,6 +775,12 @@ at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out,
     gemm_bt_kernel<false, false, true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
         C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr);
+  else if (use_db)
+    gemm_bt_kernel<false, false, false, false, true>
+        <<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+            (const bf16*)Ac.const_data_ptr(),
+            (const bf16*)Bc.const_data_ptr(), C.data_ptr(), M, N, K, nbm,
+            nbn, sm, nullptr);
   else
     gemm_bt_kernel<false, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
@@ -730,12 +806,25 @@ std::vector<at::Tensor> gemm_bt_stats(at::Tensor A, at::Tensor B) {
   auto stats = at::empty({nbm, 2 * N}, Ac.options().dtype(at::kFloat));
   StrideMap sm{0, 0, 0, 0, 1};
   auto stream = at::cuda::getCurrentCUDAStream();
+  const bool use_db = gemm_db_enabled(K);
   if (N % 256 == 0 && K <= 256 && (long)nbm * (nbn / 2) >= 256) {
     int nbn2 = nbn / 2;
-    gemm_bt_n2_kernel<<<nbm * nbn2, GEMM_TPB, 0, stream>>>(
-        (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-        (bf16*)C.data_ptr(), M, N, K, nbm, nbn2, stats.data_ptr<float>(),
-        nbm);
+    if (use_db)
+      gemm_bt_n2_kernel<true><<<nbm * nbn2, GEMM_TPB, 0, stream>>>(
+          (const bf16*)Ac.const_data_ptr(),
+          (const bf16*)Bc.const_data_ptr(), (bf16*)C.data_ptr(), M, N, K,
+          nbm, nbn2, stats.data_ptr<float>(), nbm);
+    else
+      gemm_bt_n2_kernel<<<nbm * nbn2, GEMM_TPB, 0, stream>>>(
+          (const bf16*)Ac.const_data_ptr(),
+          (const bf16*)Bc.const_data_ptr(), (bf16*)C.data_ptr(), M, N, K,
+          nbm, nbn2, stats.data_ptr<float>(), nbm);
+  } else if (use_db) {
+    gemm_bt_kernel<false, false, false, false, true>
+        <<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+            (const bf16*)Ac.const_data_ptr(),
+            (const bf16*)Bc.const_data_ptr(), C.data_ptr(), M, N, K, nbm,
+            nbn, sm, stats.data_ptr<float>());
   } else {
     gemm_bt_kernel<false, false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
@@ -762,9 +851,16 @@ at::Tensor gemm_bt_strided(at::Tensor A, at::Tensor B, long Nn, long H,
   int nbm = (int)((M + BM - 1) / BM), nbn = (int)((N + BN - 1) / BN);
   StrideMap sm{(int)H, (int)W, (int)Hout, (int)Wout, (int)stride};
   auto stream = at::cuda::getCurrentCUDAStream();
-  gemm_bt_kernel<false, true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
-      (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
-      C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr);
+  if (gemm_db_enabled(K))
+    gemm_bt_kernel<false, true, false, false, true>
+        <<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+            (const bf16*)Ac.const_data_ptr(),
+            (const bf16*)Bc.const_data_ptr(), C.data_ptr(), M, N, K, nbm,
+            nbn, sm, nullptr);
+  else
+    gemm_bt_kernel<false, true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+        (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
+        C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr);
   CHECK_CUDA_OK();
   return C;
 }
