@@ -168,14 +168,17 @@ __device__ __forceinline__ void stage_im2col8(
 
 // DGRAD instantiation computes dX[m, cin] from gathered dY rows and the
 // permuted weight; operand roles mirror fwd exactly.
-template <bool DGRAD = false, bool C8 = false>
+// DB: double-buffered K-loop (gemm.hip pattern — stage ks+1's tiles while
+// ks computes, one barrier per chunk), AMDTRAIN_STEM_DB=0 reverts.
+template <bool DGRAD = false, bool C8 = false, bool DB = false>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 conv_generic_fwd_kernel(const bf16* __restrict__ x,
                         const bf16* __restrict__ W2, bf16* __restrict__ Y,
                         long M, int Cout, StemGeom g, int nbm, int nbn,
                         const bf16* __restrict__ zp) {
-  __shared__ bf16 As[128 * BK];
-  __shared__ bf16 Bs[128 * BK];
+  __shared__ bf16 SMEM[(DB ? 2 : 1) * 2 * 128 * BK];
+  bf16* const As = SMEM;
+  bf16* const Bs = SMEM + 128 * BK;
   const int bid = blockIdx.x;
   const int bm = bid / nbn, bn = bid % nbn;
   const long m0 = (long)bm * 128, n0 = (long)bn * 128;
@@ -200,27 +203,46 @@ conv_generic_fwd_kernel(const bf16* __restrict__ x,
     uc[rnd] = stem_decode<DGRAD>(valid[rnd] ? m : M - 1, g);
   }
 
-  for (int ks = 0; ks < g.Kpad / BK; ++ks) {
-    __syncthreads();
+  const int ksteps = g.Kpad / BK;
+  auto stage = [&](int ks, bf16* as) {
     if (C8)
-      stage_im2col8(x, uc, valid, ks * BK, g, zp, As);
+      stage_im2col8(x, uc, valid, ks * BK, g, zp, as);
     else
-      stage_im2col<DGRAD>(x, uc, valid, ks * BK, g, inv_cin, inv_kw, As);
-    stage_rows(W2, g.Kpad, n0, Cout, ks * BK, Bs);
-    __syncthreads();
+      stage_im2col<DGRAD>(x, uc, valid, ks * BK, g, inv_cin, inv_kw, as);
+    stage_rows(W2, g.Kpad, n0, Cout, ks * BK, as + 128 * BK);
+  };
+  auto compute = [&](const bf16* as) {
+    const bf16* bs = as + 128 * BK;
     bf16x8 a[4], b[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
-      a[i] = *(const bf16x8*)&As[(wm + i * 16 + fr) * BK + fq * 8];
+      a[i] = *(const bf16x8*)&as[(wm + i * 16 + fr) * BK + fq * 8];
 #pragma unroll
     for (int j = 0; j < 4; ++j)
-      b[j] = *(const bf16x8*)&Bs[(wn + j * 16 + fr) * BK + fq * 8];
+      b[j] = *(const bf16x8*)&bs[(wn + j * 16 + fr) * BK + fq * 8];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
 #pragma unroll
       for (int j = 0; j < 4; ++j)
         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a[i], b[j], acc[i][j], 0, 0, 0);
+  };
+  if (DB) {
+    constexpr int HB = 2 * 128 * BK;
+    if (ksteps > 0) stage(0, SMEM);
+    for (int ks = 0; ks < ksteps; ++ks) {
+      bf16* const as = SMEM + (ks & 1) * HB;
+      __syncthreads();  // implicit vmcnt(0) drains this chunk's DMA
+      if (ks + 1 < ksteps) stage(ks + 1, SMEM + ((ks + 1) & 1) * HB);
+      compute(as);
+    }
+  } else {
+    for (int ks = 0; ks < ksteps; ++ks) {
+      __syncthreads();
+      stage(ks, SMEM);
+      __syncthreads();
+      compute(SMEM);
+    }
   }
 #pragma unroll
   for (int i = 0; i < 4; ++i)
@@ -371,6 +393,16 @@ static at::Tensor stem_zero_page(const at::Tensor& like) {
 
 }  // namespace
 
+
+// AMDTRAIN_STEM_DB=0 reverts to the single-buffer K-loop
+static bool stem_db_enabled() {
+  static const bool v = []() {
+    const char* e = std::getenv("AMDTRAIN_STEM_DB");
+    return !(e && e[0] == '0');
+  }();
+  return v;
+}
+
 // x2d: [N*H*W, Cin] bf16 NHWC rows; w2: [Cout, Kpad] (host-padded).
 // Cin == 8 (channel-padded stem) takes the vectorized tap-gather staging.
 at::Tensor conv_generic_fwd(at::Tensor x2d, long Nn, long H, long W,
@@ -385,11 +417,26 @@ at::Tensor conv_generic_fwd(at::Tensor x2d, long Nn, long H, long W,
   int nbm = (int)((M + 127) / 128), nbn = (int)((Cout + 127) / 128);
   auto stream = at::cuda::getCurrentCUDAStream();
   auto zp = stem_zero_page(x2d);
-  if (Cin == 8 && g.Kpad == (KH * KW * 8 + 31) / 32 * 32)
-    conv_generic_fwd_kernel<false, true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
-        (const bf16*)x2d.const_data_ptr(), (const bf16*)w2.const_data_ptr(),
-        (bf16*)y.data_ptr(), M, (int)Cout, g, nbm, nbn,
-        (const bf16*)zp.const_data_ptr());
+  const bool db = stem_db_enabled();
+  if (Cin == 8 && g.Kpad == (KH * KW * 8 + 31) / 32 * 32) {
+    if (db)
+      conv_generic_fwd_kernel<false, true, true>
+          <<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+              (const bf16*)x2d.const_data_ptr(),
+              (const bf16*)w2.const_data_ptr(), (bf16*)y.data_ptr(), M,
+              (int)Cout, g, nbm, nbn, (const bf16*)zp.const_data_ptr());
+    else
+      conv_generic_fwd_kernel<false, true>
+          <<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+              (const bf16*)x2d.const_data_ptr(),
+              (const bf16*)w2.const_data_ptr(), (bf16*)y.data_ptr(), M,
+              (int)Cout, g, nbm, nbn, (const bf16*)zp.const_data_ptr());
+  } else if (db)
+    conv_generic_fwd_kernel<false, false, true>
+        <<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+            (const bf16*)x2d.const_data_ptr(),
+            (const bf16*)w2.const_data_ptr(), (bf16*)y.data_ptr(), M,
+            (int)Cout, g, nbm, nbn, (const bf16*)zp.const_data_ptr());
   else
     conv_generic_fwd_kernel<false><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)x2d.const_data_ptr(), (const bf16*)w2.const_data_ptr(),
@@ -415,10 +462,17 @@ at::Tensor conv_generic_dgrad(at::Tensor dy2d, at::Tensor w2p, long Nn,
   int nbm = (int)((M + 127) / 128), nbn = (int)((Cin + 127) / 128);
   auto stream = at::cuda::getCurrentCUDAStream();
   auto zp = stem_zero_page(dy2d);
-  conv_generic_fwd_kernel<true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
-      (const bf16*)dy2d.const_data_ptr(), (const bf16*)w2p.const_data_ptr(),
-      (bf16*)dx.data_ptr(), M, (int)Cin, g, nbm, nbn,
-      (const bf16*)zp.const_data_ptr());
+  if (stem_db_enabled())
+    conv_generic_fwd_kernel<true, false, true>
+        <<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+            (const bf16*)dy2d.const_data_ptr(),
+            (const bf16*)w2p.const_data_ptr(), (bf16*)dx.data_ptr(), M,
+            (int)Cin, g, nbm, nbn, (const bf16*)zp.const_data_ptr());
+  else
+    conv_generic_fwd_kernel<true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+        (const bf16*)dy2d.const_data_ptr(),
+        (const bf16*)w2p.const_data_ptr(), (bf16*)dx.data_ptr(), M,
+        (int)Cin, g, nbm, nbn, (const bf16*)zp.const_data_ptr());
   CHECK_CUDA_OK();
   return dx;
 }
