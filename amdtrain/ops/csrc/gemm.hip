@@ -90,6 +90,14 @@ __device__ __forceinline__ void stage_tile_rows(
 // helps) -> ON by default for K <= GEMM_DB_KMAX, AMDTRAIN_GEMM_DB=0/1
 // forces either path.
 constexpr long GEMM_DB_KMAX = 1024;
+// AMDTRAIN_GEMM_ADIR=1: opt-in direct-from-global A fragments (see kernel)
+inline bool gemm_adir_enabled() {
+  static const bool v = []() {
+    const char* e = std::getenv("AMDTRAIN_GEMM_ADIR");
+    return e && e[0] == '1';
+  }();
+  return v;
+}
 inline bool gemm_db_enabled(long K) {
   static const int v = []() {
     const char* e = std::getenv("AMDTRAIN_GEMM_DB");
@@ -185,15 +193,22 @@ __device__ __forceinline__ void epilogue_stats(
 // overlap instead of relying on the 2-blocks/CU interleave drifting into
 // opposite phase).  The barrier's implicit vmcnt(0) wait is what drains
 // the global_load_lds queue, exactly as in the single-buffer path.
+// ADIR=1: A fragments are read DIRECTLY from global memory into VGPRs —
+// the mfma A fragment is 16 contiguous bytes of one A row (lane l reads
+// A[row=l&15][k=(l>>4)*8..+7]), so no LDS transit is needed at all: the
+// A-side DMA and half the LDS footprint disappear; the second wave on
+// the same rows hits L1.  A/B experiment for the streaming shapes
+// (AMDTRAIN_GEMM_ADIR).  Plain (non-strided, non-LDSE) path only.
 template <bool F32OUT, bool STRIDED, bool NT = false, bool LDSE = false,
-          bool DB = false>
+          bool DB = false, bool ADIR = false>
 __global__ void __launch_bounds__(GEMM_TPB, 2)
 gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                void* __restrict__ C, long M, long N, long K, int nbm,
                int nbn, StrideMap sm, float* __restrict__ stats) {
-  __shared__ bf16 SMEM[(DB ? 2 : 1) * (BM + BN) * BK];  // As | Bs
+  __shared__ bf16 SMEM[(DB ? 2 : 1) *
+                       ((ADIR ? 0 : BM) + BN) * BK];  // [As |] Bs
   bf16* const As = SMEM;    // (contiguous: the LDSE epilogue reuses 16 KB)
-  bf16* const Bs = SMEM + BM * BK;
+  bf16* const Bs = SMEM + (ADIR ? 0 : BM * BK);
 
   const int nwg = nbm * nbn;
   const int bid = xcd_swizzle(blockIdx.x, nwg);
@@ -224,20 +239,34 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
     }
   }
 
+  // ADIR: per-lane clamped global row pointers for the 4 A fragments
+  const bf16* Ap[4];
+  if (ADIR) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      long row = m0 + wm + i * 16 + fr;
+      if (row >= M) row = M - 1;  // finite garbage, epilogue masks it
+      Ap[i] = A + row * K + fq * 8;
+    }
+  }
   const long ksteps = K / BK;
   auto stage = [&](long kt, bf16* as) {
-    if (STRIDED)
-      stage_tile_rows(A, K, arow, kt * BK, as);
-    else
-      stage_tile_128x32(A, K, m0, M, kt * BK, as);
-    stage_tile_128x32(B, K, n0, N, kt * BK, as + BM * BK);
+    if (!ADIR) {
+      if (STRIDED)
+        stage_tile_rows(A, K, arow, kt * BK, as);
+      else
+        stage_tile_128x32(A, K, m0, M, kt * BK, as);
+    }
+    stage_tile_128x32(B, K, n0, N, kt * BK,
+                      as + (ADIR ? 0 : BM * BK));
   };
-  auto compute = [&](const bf16* as) {
-    const bf16* bs = as + BM * BK;
+  auto compute = [&](const bf16* as, long kt) {
+    const bf16* bs = as + (ADIR ? 0 : BM * BK);
     bf16x8 a[4], b[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
-      a[i] = *(const bf16x8*)&as[(wm + i * 16 + fr) * BK + fq * 8];
+      a[i] = ADIR ? *(const bf16x8*)(Ap[i] + kt * BK)
+                  : *(const bf16x8*)&as[(wm + i * 16 + fr) * BK + fq * 8];
 #pragma unroll
     for (int j = 0; j < 4; ++j)
       b[j] = *(const bf16x8*)&bs[(wn + j * 16 + fr) * BK + fq * 8];
@@ -249,7 +278,7 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
             a[i], b[j], acc[i][j], 0, 0, 0);
   };
   if (DB) {
-    constexpr int HB = (BM + BN) * BK;
+    constexpr int HB = ((ADIR ? 0 : BM) + BN) * BK;
     if (ksteps > 0) stage(0, SMEM);
     for (long kt = 0; kt < ksteps; ++kt) {
       bf16* const as = SMEM + (kt & 1) * HB;
@@ -257,14 +286,14 @@ gemm_bt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                         // vmcnt(0)); all waves are past reading the
                         // other buffer, which kt+1 overwrites below
       if (kt + 1 < ksteps) stage(kt + 1, SMEM + ((kt + 1) & 1) * HB);
-      compute(as);
+      compute(as, kt);
     }
   } else {
     for (long kt = 0; kt < ksteps; ++kt) {
       __syncthreads();  // previous compute done before overwriting LDS
       stage(kt, SMEM);
       __syncthreads();  // barrier drains the global_load_lds queue
-      compute(SMEM);
+      compute(SMEM, kt);
     }
   }
 
@@ -775,6 +804,12 @@ at::Tensor gemm_bt(at::Tensor A, at::Tensor B, bool f32_out,
     gemm_bt_kernel<false, false, true><<<nbm * nbn, GEMM_TPB, 0, stream>>>(
         (const bf16*)Ac.const_data_ptr(), (const bf16*)Bc.const_data_ptr(),
         C.data_ptr(), M, N, K, nbm, nbn, sm, nullptr);
+  else if (gemm_adir_enabled())
+    gemm_bt_kernel<false, false, false, false, true, true>
+        <<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+            (const bf16*)Ac.const_data_ptr(),
+            (const bf16*)Bc.const_data_ptr(), C.data_ptr(), M, N, K, nbm,
+            nbn, sm, nullptr);
   else if (use_db)
     gemm_bt_kernel<false, false, false, false, true>
         <<<nbm * nbn, GEMM_TPB, 0, stream>>>(
@@ -819,6 +854,12 @@ std::vector<at::Tensor> gemm_bt_stats(at::Tensor A, at::Tensor B) {
           (const bf16*)Ac.const_data_ptr(),
           (const bf16*)Bc.const_data_ptr(), (bf16*)C.data_ptr(), M, N, K,
           nbm, nbn2, stats.data_ptr<float>(), nbm);
+  } else if (gemm_adir_enabled()) {
+    gemm_bt_kernel<false, false, false, false, true, true>
+        <<<nbm * nbn, GEMM_TPB, 0, stream>>>(
+            (const bf16*)Ac.const_data_ptr(),
+            (const bf16*)Bc.const_data_ptr(), C.data_ptr(), M, N, K, nbm,
+            nbn, sm, stats.data_ptr<float>());
   } else if (use_db) {
     gemm_bt_kernel<false, false, false, false, true>
         <<<nbm * nbn, GEMM_TPB, 0, stream>>>(
