@@ -21,6 +21,6 @@ Layout (mirrors SURVEY.md §1's logical layers):
   amdtrain.utils     — meters, metrics, LR schedule, checkpointing (aux)
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 from . import utils  # noqa: F401
