@@ -74,3 +74,30 @@ def test_torchrun_bench_contract_cpu(tmp_path):
     assert d["unit"] == "images/sec" and d["higher_is_better"] is True
     assert d["config"]["global_batch"] == 4
     assert d["scaling"] == "weak" and d["vs_baseline"] is not None
+
+
+def test_torchrun_bench_contract_styles_cpu(tmp_path):
+    """The bench's --style apex/horovod variants must satisfy the same
+    JSON contract as ddp (the styles share the timing harness; this
+    guards the style-flag plumbing the driver never exercises)."""
+    import json
+    for port, style in ((29618, "horovod"), (29619, "apex")):
+        cmd = [sys.executable, "-m", "torch.distributed.run",
+               "--nnodes=1", "--nproc-per-node=2",
+               "--master-addr", "127.0.0.1", "--master-port", str(port),
+               os.path.join(ROOT, "bench.py"), "--gpus", "2", "--steps",
+               "2", "--warmup", "1", "--batch-per-gpu", "2", "--arch",
+               "resnet18", "--image-size", "32", "--dtype", "fp32",
+               "--style", style]
+        env = dict(os.environ, AMDTRAIN_BENCH_CPU="1",
+                   AMDTRAIN_DISABLE_EXT="1", AMDTRAIN_ALLOW_EAGER="1",
+                   PYTHONPATH=ROOT + os.pathsep +
+                   os.environ.get("PYTHONPATH", ""))
+        r = subprocess.run(cmd, cwd=str(tmp_path), env=env,
+                           capture_output=True, text=True, timeout=420)
+        assert r.returncode == 0, style + ": " + r.stdout[-1500:] \
+            + r.stderr[-1500:]
+        line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+        d = json.loads(line)
+        assert d["config"]["style"] == style
+        assert d["n_gpus"] == 2 and d["config"]["global_batch"] == 4
