@@ -111,3 +111,15 @@ def test_epoch_timer_csv(tmp_path):
     lines = path.read_text().strip().splitlines()
     assert lines[0] == "epoch,seconds"
     assert len(lines) == 3
+
+
+def test_profile_steps_cpu(tmp_path):
+    """profile_steps wraps torch.profiler and writes a kernel table
+    (CPU activities here; CUDA rows appear on hardware)."""
+    import torch
+    from amdtrain.utils.profiling import profile_steps
+    out = tmp_path / "prof.txt"
+    with profile_steps(str(out)):
+        torch.randn(64, 64) @ torch.randn(64, 64)
+    text = out.read_text()
+    assert "Name" in text and "CPU" in text
