@@ -59,3 +59,19 @@ def test_local_rank_flag_on_ddp_entrypoints():
     assert a.local_rank == 3
     a = parse_args(["--local-rank", "2"])  # new-style spelling
     assert a.local_rank == 2
+
+
+def test_start_sh_modules_exist():
+    """Every `-m amdtrain.cli.X` in scripts/start.sh must be a real,
+    importable entrypoint (guards doc rot in the canonical launch lines,
+    reference start.sh:1-5)."""
+    import importlib
+    import os
+    import re
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    text = open(os.path.join(root, "scripts", "start.sh")).read()
+    mods = set(re.findall(r"-m (amdtrain\.cli\.\w+)", text))
+    assert len(mods) == 6, mods
+    for m in sorted(mods):
+        mod = importlib.import_module(m)
+        assert hasattr(mod, "main"), m
